@@ -1,0 +1,48 @@
+"""Vanilla Policy Gradient (REINFORCE with GAE baseline).
+
+API parity: reference src/rl_replicas/algorithms/vpg.py:29-244.
+Policy loss = -mean(log pi(a|s) * A) with ONE policy gradient step per
+epoch and `num_value_gradients` (default 80) value-function steps.
+"""
+from __future__ import annotations
+
+from typing import Dict
+
+import torch
+from torch import Tensor
+
+from rl_replicas_amd.algorithms.on_policy import OnPolicyAlgorithm
+from rl_replicas_amd.policies import Policy
+from rl_replicas_amd.samplers import Sampler
+from rl_replicas_amd.value_function import ValueFunction
+
+
+class VPG(OnPolicyAlgorithm):
+    def __init__(
+        self,
+        policy: Policy,
+        value_function: ValueFunction,
+        env,
+        sampler: Sampler,
+        gamma: float = 0.99,
+        gae_lambda: float = 0.97,
+        num_value_gradients: int = 80,
+    ) -> None:
+        super().__init__(policy, value_function, env, sampler, gamma, gae_lambda, num_value_gradients)
+
+    def _update_policy(self, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[str, float]:
+        diagnostics = self._policy_diagnostics(obs, actions)
+
+        loss = self.compute_policy_loss(obs, actions, advantages)
+        self.policy.optimizer.zero_grad()
+        loss.backward()
+        self._all_reduce_gradients(self.policy)
+        self.policy.optimizer.step()
+
+        return {"policy/loss": float(loss.detach()), **diagnostics}
+
+    def compute_policy_loss(self, observations: Tensor, actions: Tensor, advantages: Tensor) -> Tensor:
+        """-E[log pi(a|s) * A]  (reference vpg.py:200-203)."""
+        dist = self.policy(observations)
+        log_probs = dist.log_prob(actions)
+        return -torch.mean(log_probs * advantages)

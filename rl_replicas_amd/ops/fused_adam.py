@@ -1,0 +1,68 @@
+"""Fused multi-tensor Adam for ROCm.
+
+One HIP kernel updates every parameter of the model per step instead
+of the per-tensor loop of torch.optim.Adam (reference uses plain
+torch.optim.Adam, e.g. benchmarks/run_vpg.py Adam(3e-4)).  State
+layout (`exp_avg`, `exp_avg_sq`, `step`) is identical to
+torch.optim.Adam so checkpoints interoperate
+(`*_optimizer_state_dict` keys in the reference's model.pt schema).
+"""
+from __future__ import annotations
+
+from typing import List
+
+import torch
+from torch import Tensor
+
+
+class FusedAdam(torch.optim.Adam):
+    """torch.optim.Adam with a fused multi-tensor HIP step."""
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+
+        from rl_replicas_amd import ops
+
+        ext = ops._load_extension()
+        if ext is None:  # pragma: no cover - make_adam gates on hip_available
+            return super().step()
+
+        for group in self.param_groups:
+            params: List[Tensor] = []
+            grads: List[Tensor] = []
+            exp_avgs: List[Tensor] = []
+            exp_avg_sqs: List[Tensor] = []
+            steps: List[Tensor] = []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = torch.zeros((), dtype=torch.float32, device=p.device)
+                    state["exp_avg"] = torch.zeros_like(p)
+                    state["exp_avg_sq"] = torch.zeros_like(p)
+                params.append(p.data)
+                grads.append(p.grad.data)
+                exp_avgs.append(state["exp_avg"])
+                exp_avg_sqs.append(state["exp_avg_sq"])
+                steps.append(state["step"])
+            if not params:
+                continue
+            beta1, beta2 = group["betas"]
+            ext.fused_adam_(
+                params,
+                grads,
+                exp_avgs,
+                exp_avg_sqs,
+                steps,
+                float(group["lr"]),
+                float(beta1),
+                float(beta2),
+                float(group["eps"]),
+                float(group.get("weight_decay", 0.0)),
+            )
+        return loss

@@ -1,0 +1,226 @@
+"""TRPO's trust-region optimizer: CG solve + backtracking line search.
+
+Behavioral parity with the reference `ConjugateGradientOptimizer`
+(reference: src/rl_replicas/optimizers/conjugate_gradient_optimizer.py
+:24-265), including its exact robustness semantics:
+NaN step directions zeroed, NaN step size -> 1.0, destructive in-place
+parameter writes during the line search with full rollback on
+rejection, and the strict (>=) rejection re-check after the loop.
+
+MI355X notes: the Fisher-vector product is two autograd passes through
+the fused-MLP HIP ops (each custom op implements double backward via
+its composite autograd definition); the flat parameter/direction
+vectors stay resident on device across all CG iterations (dot/axpy run
+as device ops; for the tiny models here the host loop cost is the
+launch latency, which HIP-graph capture of the CG body can amortize).
+"""
+from __future__ import annotations
+
+import logging
+from typing import Callable, Iterable, List, Tuple
+
+import torch
+from torch import Tensor
+from torch.optim import Optimizer
+
+try:
+    from typing_extensions import TypedDict
+except ImportError:  # pragma: no cover
+    from typing import TypedDict  # type: ignore
+
+logger = logging.getLogger(__name__)
+
+
+State = TypedDict(
+    "State",
+    {
+        "max_constraint": float,
+        "n_conjugate_gradients": int,
+        "max_backtracks": int,
+        "backtrack_ratio": float,
+        "hvp_damping_coefficient": float,
+    },
+)
+
+
+def _flatten(tensors: Iterable[Tensor]) -> Tensor:
+    return torch.cat([t.reshape(-1) for t in tensors])
+
+
+class ConjugateGradientOptimizer(Optimizer):
+    """Constrained step: x = H^-1 g via CG, scaled to the KL trust region.
+
+    H is the Hessian of the constraint (KL), g the loss gradient.
+    """
+
+    def __init__(
+        self,
+        params: Iterable[Tensor],
+        max_constraint: float = 0.01,
+        n_conjugate_gradients: int = 10,
+        max_backtracks: int = 15,
+        backtrack_ratio: float = 0.8,
+        hvp_damping_coefficient: float = 1e-5,
+    ):
+        super().__init__(params, {})
+        self.max_constraint = max_constraint
+        self.n_conjugate_gradients = n_conjugate_gradients
+        self.max_backtracks = max_backtracks
+        self.backtrack_ratio = backtrack_ratio
+        self.hvp_damping_coefficient = hvp_damping_coefficient
+
+    # ------------------------------------------------------------------
+    def step(self, loss_function: Callable, kl_divergence_function: Callable) -> None:  # type: ignore[override]
+        """One trust-region update.
+
+        Caller must have populated `.grad` of the params with the loss
+        gradient (reference trpo.py:236-240 does `loss.backward()` then
+        `optimizer.step(loss_fn, kl_fn)`).
+        """
+        params: List[Tensor] = [
+            p for group in self.param_groups for p in group["params"] if p.grad is not None
+        ]
+        loss_grad = _flatten([p.grad for p in params]).detach()
+
+        hvp = self._make_fisher_vector_product(kl_divergence_function, params)
+        direction = self._conjugate_gradient(hvp, loss_grad)
+        # NaN guard on the direction (reference :83)
+        direction = torch.nan_to_num(direction, nan=0.0)
+
+        # beta = sqrt(2*delta / (d^T H d)) (reference :86-90)
+        quad = torch.dot(direction, hvp(direction)) + 1e-8
+        step_size = float(torch.sqrt(2.0 * self.max_constraint / quad))
+        if step_size != step_size:  # NaN guard (reference :92-94)
+            step_size = 1.0
+
+        self._backtracking_line_search(
+            params, step_size * direction, loss_function, kl_divergence_function
+        )
+
+    # ------------------------------------------------------------------
+    def _make_fisher_vector_product(
+        self, kl_function: Callable, params: List[Tensor]
+    ) -> Callable[[Tensor], Tensor]:
+        """FVP via double backward: v -> H v + damping*v.
+
+        The KL gradient graph is built ONCE (create_graph=True); each
+        call does one backward through it (reference :133-167).
+        """
+        kl = kl_function()
+        kl_grads: Tuple[Tensor, ...] = torch.autograd.grad(kl, params, create_graph=True)
+        flat_kl_grad = _flatten(kl_grads)
+
+        def fvp(vector: Tensor) -> Tensor:
+            gvp = torch.dot(flat_kl_grad, vector)
+            hvp_list = torch.autograd.grad(gvp, params, retain_graph=True, allow_unused=True)
+            hvp_list = [
+                h if h is not None else torch.zeros_like(p) for h, p in zip(hvp_list, params)
+            ]
+            return _flatten(hvp_list) + self.hvp_damping_coefficient * vector
+
+        return fvp
+
+    # ------------------------------------------------------------------
+    def _conjugate_gradient(
+        self, hvp: Callable[[Tensor], Tensor], b: Tensor, residual_tol: float = 1e-10
+    ) -> Tensor:
+        """Solve H x = b (standard CG; reference :169-202).
+
+        All vectors stay on b's device; the host loop only reads the
+        scalar residual for the early-exit test.
+        """
+        x = torch.zeros_like(b)
+        r = b.clone()
+        p = b.clone()
+        r_dot_r = torch.dot(r, r)
+        for _ in range(self.n_conjugate_gradients):
+            hp = hvp(p)
+            alpha = r_dot_r / torch.dot(p, hp)
+            x = x + alpha * p
+            r = r - alpha * hp
+            new_r_dot_r = torch.dot(r, r)
+            p = r + (new_r_dot_r / r_dot_r) * p
+            r_dot_r = new_r_dot_r
+            if float(r_dot_r) < residual_tol:
+                break
+        return x
+
+    # ------------------------------------------------------------------
+    def _backtracking_line_search(
+        self,
+        params: List[Tensor],
+        descent_step: Tensor,
+        loss_function: Callable,
+        constraint_function: Callable,
+    ) -> None:
+        """Shrink the step by `backtrack_ratio` until loss improves and
+        the KL constraint holds; roll back entirely on failure
+        (reference :204-250).
+        """
+        saved = [p.detach().clone() for p in params]
+        loss_before = loss_function()
+
+        # per-parameter views of the flat step
+        numels = [p.numel() for p in params]
+        step_views = [
+            s.view(p.shape)
+            for s, p in zip(torch.split(descent_step, numels), params)
+        ]
+
+        new_loss = loss_before
+        constraint = torch.zeros((), device=descent_step.device)
+        for i in range(self.max_backtracks):
+            ratio = self.backtrack_ratio**i
+            with torch.no_grad():
+                for p, p0, s in zip(params, saved, step_views):
+                    p.data.copy_(p0 - ratio * s)
+            new_loss = loss_function()
+            constraint = constraint_function()
+            if new_loss < loss_before and constraint <= self.max_constraint:
+                break
+
+        # strict post-check: reject & roll back (reference :230-250)
+        if (
+            torch.isnan(new_loss)
+            or torch.isnan(constraint)
+            or new_loss >= loss_before
+            or constraint >= self.max_constraint
+        ):
+            logger.warning("Line search condition violated. Rejecting the step.")
+            with torch.no_grad():
+                for p, p0 in zip(params, saved):
+                    p.data.copy_(p0)
+
+    # ------------------------------------------------------------------
+    # state / resume (reference :100-131)
+    # ------------------------------------------------------------------
+    @property
+    def state(self) -> State:  # type: ignore[override]
+        return {
+            "max_constraint": self.max_constraint,
+            "n_conjugate_gradients": self.n_conjugate_gradients,
+            "max_backtracks": self.max_backtracks,
+            "backtrack_ratio": self.backtrack_ratio,
+            "hvp_damping_coefficient": self.hvp_damping_coefficient,
+        }
+
+    @state.setter
+    def state(self, state: State) -> None:
+        self.max_constraint = state.get("max_constraint", 0.01)
+        self.n_conjugate_gradients = state.get("n_conjugate_gradients", 10)
+        self.max_backtracks = state.get("max_backtracks", 15)
+        self.backtrack_ratio = state.get("backtrack_ratio", 0.8)
+        self.hvp_damping_coefficient = state.get("hvp_damping_coefficient", 1e-5)
+
+    def __setstate__(self, state: dict) -> None:
+        if "hvp_damping_coefficient" not in state["state"]:
+            logger.warning("Resuming ConjugateGradientOptimizer with lost state.")
+        self.state = state["state"]
+        self.param_groups = state["param_groups"]
+
+    def unflatten_tensor(self, flattened: Tensor, shapes: List[torch.Size]) -> List[Tensor]:
+        """Split a flat vector back into tensors of `shapes` (reference :252-265)."""
+        numels = [int(torch.Size(s).numel()) for s in shapes]
+        return [
+            chunk.view(shape) for chunk, shape in zip(torch.split(flattened, numels), shapes)
+        ]

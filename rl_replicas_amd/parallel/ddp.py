@@ -1,0 +1,166 @@
+"""Data parallelism over RCCL/xGMI.
+
+The reference has no distributed anything.  This
+module adds single-node data parallelism designed for the MI355X xGMI
+topology: one process per GPU (`torch.distributed`, backend "nccl" =
+RCCL on ROCm), each rank sampling its own env shard and all-reducing
+gradients.
+
+Design choices (xGMI-first):
+* Gradients here are KILOBYTES (policy+value ~4 K params fp32), so
+  collectives are pure latency.  We therefore send ONE fused flat
+  buffer per optimizer step (torch._utils flatten/unflatten) rather
+  than per-parameter or bucketed messages; overlap machinery would
+  only add launch latency at this size.
+* Advantage normalization must be GLOBAL to reproduce single-GPU
+  numerics: ranks all-reduce [sum, sum_sq, count] (one 3-float
+  message) and normalize with the global Bessel-corrected std.
+* PPO's early-stop KL is all-reduced (mean) so every rank takes the
+  same number of policy steps — replicas stay bitwise identical.
+* Rank failure aborts the job (no elasticity on one node; the
+  reference's only robustness is numerical — SURVEY.md §5.3).
+
+Works on CPU with the gloo backend for testing (tests/test_parallel_cpu.py).
+"""
+from __future__ import annotations
+
+import logging
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+from torch import Tensor
+from torch._utils import _flatten_dense_tensors, _unflatten_dense_tensors
+
+logger = logging.getLogger(__name__)
+
+
+def distributed_is_active() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if distributed_is_active() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if distributed_is_active() else 1
+
+
+def init_from_env(backend: Optional[str] = None) -> int:
+    """Initialize torch.distributed from torchrun env vars; returns rank.
+
+    Selects RCCL ("nccl") when a GPU is visible, gloo otherwise, binds
+    the process to its LOCAL_RANK GPU, and is a no-op outside a
+    distributed launch (WORLD_SIZE unset or 1 with no MASTER_ADDR).
+    """
+    if distributed_is_active():
+        return dist.get_rank()
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    if world_size <= 1 and "MASTER_ADDR" not in os.environ:
+        return 0
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local_rank)
+    dist.init_process_group(backend=backend)
+    logger.info(
+        "initialized process group: backend=%s rank=%d world=%d",
+        backend,
+        dist.get_rank(),
+        dist.get_world_size(),
+    )
+    return dist.get_rank()
+
+
+# ---------------------------------------------------------------------------
+def _broadcast_module(module: torch.nn.Module, src: int = 0) -> None:
+    for t in list(module.parameters()) + list(module.buffers()):
+        dist.broadcast(t.data, src=src)
+
+
+def all_reduce_gradients(module: torch.nn.Module) -> None:
+    """Average gradients across ranks as ONE fused flat message."""
+    if not distributed_is_active():
+        return
+    grads: List[Tensor] = [p.grad for p in module.parameters() if p.grad is not None]
+    if not grads:
+        return
+    flat = _flatten_dense_tensors(grads)
+    dist.all_reduce(flat)
+    flat.div_(dist.get_world_size())
+    for g, synced in zip(grads, _unflatten_dense_tensors(flat, grads)):
+        g.copy_(synced)
+
+
+def global_normalize(x: Tensor) -> Tensor:
+    """(x - mean)/std over the GLOBAL (all-rank) batch, Bessel-corrected."""
+    if not distributed_is_active():
+        from rl_replicas_amd import ops
+
+        return ops.normalize(x)
+    stats = torch.stack([x.sum(), (x * x).sum(), torch.tensor(float(x.numel()), device=x.device, dtype=x.dtype)])
+    dist.all_reduce(stats)
+    total, total_sq, count = stats[0], stats[1], stats[2]
+    mean = total / count
+    var = (total_sq - count * mean * mean) / (count - 1)
+    return (x - mean) / torch.sqrt(var)
+
+
+def all_reduce_mean_scalar(x: Tensor) -> Tensor:
+    if not distributed_is_active():
+        return x
+    x = x.clone()
+    dist.all_reduce(x)
+    return x / dist.get_world_size()
+
+
+# ---------------------------------------------------------------------------
+def enable_data_parallel(algorithm) -> None:
+    """Wire an algorithm instance for multi-rank training.
+
+    Broadcasts every module's initial parameters from rank 0 and
+    installs the DP hooks (`_all_reduce_gradients`,
+    `_normalize_advantages`, `_reduce_scalar_mean`) the algorithm
+    templates call.  The caller is responsible for per-rank sampler
+    seeds (utils.set_seed_for_rank) and for scaling batch sizes.
+    """
+    if not distributed_is_active():
+        logger.warning("enable_data_parallel called without an initialized process group; no-op")
+        return
+
+    import types
+
+    import torch.nn as nn
+
+    for name in (
+        "policy",
+        "old_policy",
+        "value_function",
+        "q_function",
+        "q_function_1",
+        "q_function_2",
+        "target_policy",
+        "target_q_function",
+        "target_q_function_1",
+        "target_q_function_2",
+    ):
+        module = getattr(algorithm, name, None)
+        if isinstance(module, nn.Module):
+            _broadcast_module(module)
+
+    def _all_reduce_gradients(self, module) -> None:
+        all_reduce_gradients(module)
+
+    def _normalize_advantages(self, advantages: Tensor) -> Tensor:
+        return global_normalize(advantages)
+
+    def _reduce_scalar_mean(self, x: Tensor) -> Tensor:
+        return all_reduce_mean_scalar(x)
+
+    algorithm._all_reduce_gradients = types.MethodType(_all_reduce_gradients, algorithm)
+    algorithm._normalize_advantages = types.MethodType(_normalize_advantages, algorithm)
+    algorithm._reduce_scalar_mean = types.MethodType(_reduce_scalar_mean, algorithm)
+    algorithm._dp_enabled = True
