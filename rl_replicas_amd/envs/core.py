@@ -68,6 +68,9 @@ class Env:
     def reset(self, *, seed: Optional[int] = None, options: Optional[dict] = None):
         if seed is not None:
             self._np_random = np.random.default_rng(seed)
+            # derive the action-space stream from the same seed so that
+            # random-policy warm-up (DDPG/TD3) is deterministic
+            self.action_space.seed(seed + 1000)
         self._elapsed_steps = 0
         obs = self._reset_b(1)[0]
         return obs, {}

@@ -43,9 +43,11 @@ class VectorEnv:
         self._elapsed = np.zeros(self.num_envs, dtype=np.int64)
 
     def seed(self, seed: Optional[int]) -> None:
+        # same seeding contract as Env.reset(seed=...): env dynamics RNG
+        # plus a derived action-space stream
         self.env._np_random = np.random.default_rng(seed)
-        if self.action_space is not None:
-            self.action_space.seed(None if seed is None else seed + 1)
+        if seed is not None:
+            self.action_space.seed(seed + 1000)
 
     def reset(self, *, seed: Optional[int] = None) -> np.ndarray:
         if seed is not None:

@@ -1,0 +1,94 @@
+"""Multi-process data-parallel tests on the gloo backend (CPU).
+
+These pin the DP semantics the MI355X RCCL path relies on
+(SURVEY.md §2.3): N-rank gradient all-reduce == single-process
+full-batch gradients, global advantage normalization == single-batch
+normalization, and rank synchronization through a short PPO run.
+"""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+
+def _init(rank: int, world: int, tmpdir: str):
+    import torch.distributed as dist
+
+    dist.init_process_group(
+        backend="gloo",
+        init_method=f"file://{tmpdir}/pg_init",
+        rank=rank,
+        world_size=world,
+    )
+    return dist
+
+
+def _worker_allreduce(rank: int, world: int, tmpdir: str):
+    dist = _init(rank, world, tmpdir)
+    torch.manual_seed(0)  # identical model on all ranks
+    from rl_replicas_amd.parallel import all_reduce_gradients, global_normalize
+
+    net = torch.nn.Linear(4, 3)
+    full_x = torch.randn(8, 4, generator=torch.Generator().manual_seed(42))
+    full_y = torch.randn(8, 3, generator=torch.Generator().manual_seed(43))
+    # single-process oracle gradient on the full batch
+    loss_full = torch.nn.functional.mse_loss(net(full_x), full_y)
+    oracle = torch.autograd.grad(loss_full, list(net.parameters()))
+
+    # each rank computes grads on its half, then all-reduces
+    shard = slice(rank * 4, (rank + 1) * 4)
+    loss = torch.nn.functional.mse_loss(net(full_x[shard]), full_y[shard])
+    net.zero_grad()
+    loss.backward()
+    all_reduce_gradients(net)
+    for p, g_oracle in zip(net.parameters(), oracle):
+        torch.testing.assert_close(p.grad, g_oracle, rtol=1e-5, atol=1e-6)
+
+    # global normalization == normalizing the concatenated vector
+    full_v = torch.randn(10, generator=torch.Generator().manual_seed(7))
+    local = full_v[rank * 5 : (rank + 1) * 5]
+    got = global_normalize(local)
+    expected = ((full_v - full_v.mean()) / full_v.std())[rank * 5 : (rank + 1) * 5]
+    torch.testing.assert_close(got, expected, rtol=1e-5, atol=1e-6)
+    dist.destroy_process_group()
+
+
+def _worker_ppo_sync(rank: int, world: int, tmpdir: str):
+    dist = _init(rank, world, tmpdir)
+    import torch.nn as nn
+
+    from rl_replicas_amd import envs
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.parallel import enable_data_parallel
+    from rl_replicas_amd.policies import CategoricalPolicy
+    from rl_replicas_amd.samplers import BatchSampler
+    from rl_replicas_amd.utils import set_seed_for_rank
+    from rl_replicas_amd.value_function import ValueFunction
+
+    set_seed_for_rank(0, rank)  # different sampling streams per rank
+    env = envs.make("CartPole-v1")
+    pnet = MLP([4, 16, 2])
+    policy = CategoricalPolicy(pnet, torch.optim.Adam(pnet.parameters(), lr=3e-4))
+    vnet = MLP([4, 16, 1])
+    vf = ValueFunction(vnet, torch.optim.Adam(vnet.parameters(), lr=1e-3))
+    model = PPO(policy, vf, env, BatchSampler(env, seed=100 + rank))
+    enable_data_parallel(model)
+    model.learn(num_epochs=2, batch_size=200, output_dir=os.path.join(tmpdir, "out"))
+
+    # replicas must be bitwise identical after DP training
+    import torch.distributed as tdist
+
+    for p in list(policy.parameters()) + list(vf.parameters()):
+        ref = p.detach().clone()
+        tdist.broadcast(ref, src=0)
+        assert torch.equal(ref, p.detach()), "rank divergence detected"
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize("worker", [_worker_allreduce, _worker_ppo_sync])
+def test_two_rank_gloo(worker, tmp_path):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    mp.spawn(worker, args=(2, str(tmp_path)), nprocs=2, join=True)

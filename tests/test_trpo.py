@@ -1,0 +1,43 @@
+"""TRPO integration tests (protocol of reference tests/test_trpo.py)."""
+import numpy as np
+import torch
+import torch.nn as nn
+
+from rl_replicas_amd import envs
+from rl_replicas_amd.algorithms import TRPO
+from rl_replicas_amd.evaluator import Evaluator
+from rl_replicas_amd.networks import MLP
+from rl_replicas_amd.optimizers import ConjugateGradientOptimizer
+from rl_replicas_amd.policies import CategoricalPolicy, GaussianPolicy
+from rl_replicas_amd.samplers import BatchSampler
+from rl_replicas_amd.value_function import ValueFunction
+
+
+def test_trpo_cartpole_learns(tmp_path):
+    env = envs.make("CartPole-v1")
+    obs_dim = env.observation_space.shape[0]
+    pnet = MLP([obs_dim, 64, 32, env.action_space.n])
+    policy = CategoricalPolicy(pnet, ConjugateGradientOptimizer(pnet.parameters()))
+    vnet = MLP([obs_dim, 64, 32, 1])
+    vf = ValueFunction(vnet, torch.optim.Adam(vnet.parameters(), lr=1e-3))
+    model = TRPO(policy, vf, env, BatchSampler(env, seed=0))
+    model.learn(num_epochs=5, batch_size=500, output_dir=str(tmp_path))
+    returns, _ = Evaluator(seed=0).evaluate(model.policy, envs.make("CartPole-v1"), 3)
+    assert np.mean(returns) > 40.0
+
+
+def test_trpo_pendulum_learns(tmp_path):
+    env = envs.make("Pendulum-v1")
+    obs_dim = env.observation_space.shape[0]
+    act_dim = env.action_space.shape[0]
+    pnet = MLP([obs_dim, 64, 32, act_dim])
+    log_std = nn.Parameter(-0.5 * torch.ones(act_dim))
+    policy = GaussianPolicy(
+        pnet, ConjugateGradientOptimizer(list(pnet.parameters()) + [log_std]), log_std
+    )
+    vnet = MLP([obs_dim, 64, 32, 1])
+    vf = ValueFunction(vnet, torch.optim.Adam(vnet.parameters(), lr=1e-3))
+    model = TRPO(policy, vf, env, BatchSampler(env, seed=0))
+    model.learn(num_epochs=5, batch_size=500, output_dir=str(tmp_path))
+    returns, _ = Evaluator(seed=0).evaluate(model.policy, envs.make("Pendulum-v1"), 3)
+    assert np.mean(returns) > -1500.0
