@@ -1,0 +1,152 @@
+"""Flagship benchmark: PPO HalfCheetah-v4, whole-node env-steps/sec.
+
+BASELINE.json metric: "env-steps/sec (whole node) + avg return @1M
+steps, PPO HalfCheetah-v4 1/2/4/8 GPU" on synthetic-obs rollouts with
+random-init weights (no MuJoCo/network in this stack — BASELINE.md).
+
+One bench step = one full PPO epoch at the reference's published
+config (BASELINE.md "Config behind the numbers"): 4000 env steps per
+GPU (weak scaling) sampled from vectorized synthetic HalfCheetah-shaped
+envs through the policy, then GAE + advantage normalization + up to 80
+clipped-surrogate policy gradient steps (KL early stop) + 80 value
+steps — nothing skipped inside the timed region.
+
+Launch (driver contract):
+  python bench.py --gpus N --steps K --warmup W
+  # N>1 via torch.distributed.run, one rank per GPU over RCCL
+
+vs_baseline: the reference's implied serial throughput is ~946
+env-steps/s (PPO, 3M steps / 3170 s) on its CPU-only
+single-env design; that is the only published throughput number.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+import torch.nn as nn
+
+
+def build_model(device: str, num_envs: int, seed: int):
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import GaussianPolicy
+    from rl_replicas_amd.samplers import VectorSampler
+    from rl_replicas_amd.value_function import ValueFunction
+
+    obs_dim, act_dim = envs.MUJOCO_SHAPES["HalfCheetah-v4"]
+    # reference on-policy config: policy [obs,64,32,act] tanh,
+    # log_std -0.5, Adam 3e-4; value [obs,64,32,1], Adam 1e-3
+    pnet = MLP([obs_dim, 64, 32, act_dim]).to(device)
+    log_std = nn.Parameter(-0.5 * torch.ones(act_dim, device=device))
+    policy = GaussianPolicy(
+        pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+    )
+    vnet = MLP([obs_dim, 64, 32, 1]).to(device)
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+
+    venv = envs.VectorEnv("HalfCheetah-v4", num_envs=num_envs)
+    sampler = VectorSampler(venv, seed=seed)
+    model = PPO(policy, vf, venv, sampler)  # gamma .99, lambda .97, eps .2, 80/80
+    return model, sampler
+
+
+def main() -> None:
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=10)
+    parser.add_argument("--warmup", type=int, default=3)
+    parser.add_argument("--batch-per-gpu", type=int, default=4000)
+    parser.add_argument("--num-envs", type=int, default=200)
+    args = parser.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    use_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+
+    from rl_replicas_amd.parallel import enable_data_parallel, init_from_env
+
+    if world > 1:
+        init_from_env()
+
+    torch.manual_seed(1234 + rank)
+    np.random.seed(1234 + rank)
+
+    model, sampler = build_model(device, args.num_envs, seed=1234 + rank)
+    if world > 1:
+        enable_data_parallel(model)
+
+    import tempfile
+
+    model._begin_learn(tempfile.mkdtemp())
+    model.metrics_manager.stdout = False  # keep the JSON line clean
+
+    def one_epoch():
+        experience = sampler.sample(args.batch_per_gpu, model.policy)
+        model.current_total_steps += sum(experience.episode_lengths)
+        model.train(experience)
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        one_epoch()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_epoch()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks -> whole-job time
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_env_steps = world * args.batch_per_gpu * args.steps
+    value = total_env_steps / elapsed
+    if rank == 0:
+        result = {
+            "metric": "env_steps_per_sec",
+            "value": value,
+            "unit": "env-steps/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": elapsed / args.steps * 1000.0,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / 946.0,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "PPO HalfCheetah-v4 (policy MLP [17,64,32,6] tanh, value [17,64,32,1])",
+                "global_batch": world * args.batch_per_gpu,
+                "seq_len": 1000,
+                "parallelism": f"dp{world}",
+                "num_envs_per_gpu": args.num_envs,
+                "policy_grads_per_epoch": 80,
+                "value_grads_per_epoch": 80,
+                "note": "step = one PPO epoch (sample batch + full update); vs_baseline is the reference's implied ~946 env-steps/s serial CPU throughput (BASELINE.md)",
+            },
+        }
+        print(json.dumps(result))
+
+
+if __name__ == "__main__":
+    main()

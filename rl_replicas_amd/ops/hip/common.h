@@ -1,0 +1,79 @@
+// Shared device helpers for the rl_replicas_amd CDNA4 (gfx950) kernels.
+#pragma once
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+#define DEV_INLINE __device__ __forceinline__
+
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef __hip_bfloat16 bf16_t;
+
+// activation codes shared with python (ops/fused_mlp.py)
+#define ACT_IDENTITY 0
+#define ACT_TANH 1
+#define ACT_RELU 2
+
+DEV_INLINE float act_apply(int code, float x) {
+  switch (code) {
+    case ACT_TANH: return tanhf(x);
+    case ACT_RELU: return x > 0.f ? x : 0.f;
+    default: return x;
+  }
+}
+
+// derivative of the activation expressed through the POST-activation y
+// (tanh' = 1 - y^2, relu' = y > 0), so backward only needs saved outputs
+DEV_INLINE float act_grad_from_y(int code, float y) {
+  switch (code) {
+    case ACT_TANH: return 1.f - y * y;
+    case ACT_RELU: return y > 0.f ? 1.f : 0.f;
+    default: return 1.f;
+  }
+}
+
+// wave-level f32 sum (64 lanes)
+DEV_INLINE float wave_reduce_sum(float v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
+// Fused-MLP static limits (mirrored in ops/fused_mlp.py)
+#define MLP_MAX_LAYERS 5
+#define MLP_MAX_WIDTH 256
+#define MLP_ROWS 64            // rows per workgroup (4 waves x 16)
+#define MLP_LDSW (MLP_MAX_WIDTH + 4)  // padded LDS row stride (bank spread)
+
+// kernel-argument block for the fused MLP kernels (passed by value)
+struct MLPArgs {
+  const float* w[MLP_MAX_LAYERS];
+  const float* b[MLP_MAX_LAYERS];
+  float* h[MLP_MAX_LAYERS];  // saved post-activation outputs (h[L-1] = final out)
+  int dims[MLP_MAX_LAYERS + 1];
+  int acts[MLP_MAX_LAYERS];
+  int n_layers;
+  int batch;
+};
+
+// multi-tensor update argument blocks (update_kernels.hip / bindings.hip)
+#define MT_MAX_TENSORS 16
+
+struct AdamArgs {
+  float* p[MT_MAX_TENSORS];
+  float* g[MT_MAX_TENSORS];
+  float* m[MT_MAX_TENSORS];
+  float* v[MT_MAX_TENSORS];
+  float* step[MT_MAX_TENSORS];  // scalar step counters (fp32, torch-compatible)
+  int numel[MT_MAX_TENSORS];
+  int n_tensors;
+  float lr, beta1, beta2, eps, weight_decay;
+};
+
+struct PolyakArgs {
+  const float* src[MT_MAX_TENSORS];
+  float* dst[MT_MAX_TENSORS];
+  int numel[MT_MAX_TENSORS];
+  int n_tensors;
+  float rho;
+};

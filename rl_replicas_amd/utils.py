@@ -168,7 +168,11 @@ def set_seed_for_libraries(seed: int) -> None:
     torch.manual_seed(seed)
     torch.backends.cudnn.deterministic = True
     torch.backends.cudnn.benchmark = False
-    torch.use_deterministic_algorithms(True)
+    # strict on CPU (reference semantics); warn-only on ROCm where a few
+    # torch eager kernels lack a deterministic variant — our own HIP
+    # kernels are deterministic by construction (split-K workspace
+    # reductions, no atomics) and the GPU tests assert bitwise equality
+    torch.use_deterministic_algorithms(True, warn_only=torch.cuda.is_available())
 
 
 def set_seed_for_rank(seed: int, rank: Optional[int] = None) -> None:

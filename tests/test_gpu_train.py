@@ -1,0 +1,131 @@
+"""End-to-end GPU training tests (gpu-marked): each algorithm trains on
+cuda:0 through the HIP kernel path, learns on CartPole/Pendulum, and is
+bitwise-deterministic per seed."""
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+pytestmark = pytest.mark.gpu
+
+DEVICE = "cuda:0"
+
+
+@pytest.fixture(autouse=True)
+def require_hip():
+    from rl_replicas_amd import ops
+
+    assert ops.hip_available(), "HIP extension must load on GPU boxes"
+
+
+def test_ppo_cartpole_learns_on_gpu(tmp_path):
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.evaluator import Evaluator
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import CategoricalPolicy
+    from rl_replicas_amd.samplers import VectorSampler
+    from rl_replicas_amd.value_function import ValueFunction
+
+    env = envs.make("CartPole-v1")
+    venv = envs.VectorEnv("CartPole-v1", num_envs=10)
+    pnet = MLP([4, 64, 32, 2]).to(DEVICE)
+    policy = CategoricalPolicy(pnet, ops.make_adam(pnet.parameters(), lr=3e-4))
+    vnet = MLP([4, 64, 32, 1]).to(DEVICE)
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+    model = PPO(policy, vf, env, VectorSampler(venv, seed=0))
+    model.learn(num_epochs=10, batch_size=500, output_dir=str(tmp_path))
+    returns, _ = Evaluator(seed=0).evaluate(policy, envs.make("CartPole-v1"), 3)
+    assert np.mean(returns) > 35.0
+
+
+def test_td3_pendulum_gpu_with_hbm_replay(tmp_path):
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import TD3
+    from rl_replicas_amd.evaluator import Evaluator
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import DeterministicPolicy, RandomPolicy
+    from rl_replicas_amd.q_function import QFunction
+    from rl_replicas_amd.replay_buffer import ReplayBuffer
+    from rl_replicas_amd.samplers import BatchSampler
+
+    env = envs.make("Pendulum-v1")
+    pnet = MLP([3, 256, 256, 1], activation_function=nn.ReLU, output_activation_function=nn.Tanh).to(DEVICE)
+    policy = DeterministicPolicy(pnet, ops.make_adam(pnet.parameters(), lr=1e-3))
+    qs = []
+    for _ in range(2):
+        qn = MLP([4, 256, 256, 1], activation_function=nn.ReLU).to(DEVICE)
+        qs.append(QFunction(qn, ops.make_adam(qn.parameters(), lr=1e-3)))
+    model = TD3(
+        policy,
+        RandomPolicy(env.action_space),
+        qs[0],
+        qs[1],
+        env,
+        BatchSampler(env, seed=0, is_continuous=True),
+        ReplayBuffer(int(1e5), device=DEVICE),  # HBM-resident ring
+        Evaluator(seed=1),
+    )
+    model.learn(
+        num_epochs=30,
+        batch_size=50,
+        num_start_steps=500,
+        num_steps_before_update=500,
+        num_evaluation_episodes=1,
+        evaluation_interval=500,
+        output_dir=str(tmp_path),
+    )
+    # trained without error; params finite
+    for p in pnet.parameters():
+        assert torch.isfinite(p).all()
+
+
+def test_trpo_gpu_second_order_path(tmp_path):
+    """TRPO's FVP/CG runs on GPU (eager grad-enabled forwards, fused
+    no-grad forwards)."""
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import TRPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.optimizers import ConjugateGradientOptimizer
+    from rl_replicas_amd.policies import CategoricalPolicy
+    from rl_replicas_amd.samplers import VectorSampler
+    from rl_replicas_amd.value_function import ValueFunction
+
+    env = envs.make("CartPole-v1")
+    venv = envs.VectorEnv("CartPole-v1", num_envs=10)
+    pnet = MLP([4, 64, 32, 2]).to(DEVICE)
+    policy = CategoricalPolicy(pnet, ConjugateGradientOptimizer(pnet.parameters()))
+    vnet = MLP([4, 64, 32, 1]).to(DEVICE)
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+    model = TRPO(policy, vf, env, VectorSampler(venv, seed=0))
+    model.learn(num_epochs=3, batch_size=500, output_dir=str(tmp_path))
+    for p in pnet.parameters():
+        assert torch.isfinite(p).all()
+
+
+def test_gpu_determinism_same_seed(tmp_path):
+    """Same seed -> bitwise-identical parameters after training (the
+    own-stack determinism contract)."""
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import CategoricalPolicy
+    from rl_replicas_amd.samplers import VectorSampler
+    from rl_replicas_amd.utils import set_seed_for_libraries
+    from rl_replicas_amd.value_function import ValueFunction
+
+    def run(d):
+        set_seed_for_libraries(7)
+        venv = envs.VectorEnv("CartPole-v1", num_envs=10)
+        pnet = MLP([4, 32, 2]).to(DEVICE)
+        policy = CategoricalPolicy(pnet, ops.make_adam(pnet.parameters(), lr=3e-4))
+        vnet = MLP([4, 32, 1]).to(DEVICE)
+        vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+        model = PPO(policy, vf, venv, VectorSampler(venv, seed=7))
+        model.learn(num_epochs=2, batch_size=300, output_dir=str(d))
+        return [p.detach().cpu().clone() for p in pnet.parameters()]
+
+    p1 = run(tmp_path / "a")
+    p2 = run(tmp_path / "b")
+    for a, b in zip(p1, p2):
+        assert torch.equal(a, b)
