@@ -63,6 +63,7 @@ def main() -> None:
     parser.add_argument("--warmup", type=int, default=3)
     parser.add_argument("--batch-per-gpu", type=int, default=4000)
     parser.add_argument("--num-envs", type=int, default=200)
+    parser.add_argument("--phase-timing", action="store_true", help="print sample/train ms split (rank 0, stderr)")
     args = parser.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -91,10 +92,20 @@ def main() -> None:
     model._begin_learn(tempfile.mkdtemp())
     model.metrics_manager.stdout = False  # keep the JSON line clean
 
+    phase_ms = {"sample": 0.0, "train": 0.0}
+
     def one_epoch():
+        t0 = time.perf_counter()
         experience = sampler.sample(args.batch_per_gpu, model.policy)
+        if use_gpu and args.phase_timing:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
         model.current_total_steps += sum(experience.episode_lengths)
         model.train(experience)
+        if use_gpu and args.phase_timing:
+            torch.cuda.synchronize()
+        phase_ms["sample"] += (t1 - t0) * 1000.0
+        phase_ms["train"] += (time.perf_counter() - t1) * 1000.0
 
     def barrier_sync():
         if world > 1:
@@ -120,6 +131,14 @@ def main() -> None:
 
     total_env_steps = world * args.batch_per_gpu * args.steps
     value = total_env_steps / elapsed
+    if rank == 0 and args.phase_timing:
+        import sys
+
+        n = args.warmup + args.steps
+        print(
+            f"phase ms/epoch: sample={phase_ms['sample']/n:.2f} train={phase_ms['train']/n:.2f}",
+            file=sys.stderr,
+        )
     if rank == 0:
         result = {
             "metric": "env_steps_per_sec",
