@@ -133,16 +133,22 @@ class OnPolicyAlgorithm(AlgorithmBase):
 
         policy_metrics = self._update_policy(obs, actions, advantages)
 
-        value_losses: List[float] = []
-        for _ in range(self.num_value_gradients):
-            value_losses.append(self.train_value_function(obs, returns).item())
+        from rl_replicas_amd.ops import fused_onpolicy
+
+        if fused_onpolicy.value_supported(self, obs):
+            value_loss = fused_onpolicy.value_update(self, obs, returns, self.num_value_gradients)
+        else:
+            value_losses: List[float] = []
+            for _ in range(self.num_value_gradients):
+                value_losses.append(self.train_value_function(obs, returns).item())
+            value_loss = float(np.mean(value_losses))
 
         m = self.metrics_manager
         for tag, value in policy_metrics.items():
             m.record_scalar(tag, value, self.current_total_steps, tensorboard=True)
         m.record_scalar(
             "value_function/average_loss",
-            float(np.mean(value_losses)),
+            value_loss,
             self.current_total_steps,
             tensorboard=True,
         )

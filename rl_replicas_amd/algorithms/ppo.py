@@ -48,6 +48,10 @@ class PPO(OnPolicyAlgorithm):
 
     # ------------------------------------------------------------------
     def _update_policy(self, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[str, float]:
+        from rl_replicas_amd.ops import fused_onpolicy
+
+        if fused_onpolicy.supported(self.policy, obs):
+            return fused_onpolicy.ppo_update(self, obs, actions, advantages)
         # pre-update diagnostics (reference ppo.py:163-170)
         diagnostics = self._policy_diagnostics(obs, actions)
         with torch.no_grad():

@@ -31,6 +31,10 @@ class VPG(OnPolicyAlgorithm):
         super().__init__(policy, value_function, env, sampler, gamma, gae_lambda, num_value_gradients)
 
     def _update_policy(self, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[str, float]:
+        from rl_replicas_amd.ops import fused_onpolicy
+
+        if fused_onpolicy.supported(self.policy, obs):
+            return fused_onpolicy.vpg_update(self, obs, actions, advantages)
         diagnostics = self._policy_diagnostics(obs, actions)
 
         loss = self.compute_policy_loss(obs, actions, advantages)

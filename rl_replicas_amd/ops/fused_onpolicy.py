@@ -1,0 +1,230 @@
+"""Fully-fused on-policy update loops (GPU hot path).
+
+On the GPU the PPO/VPG policy-update loop and the value-function loop
+bypass torch autograd entirely: per iteration
+    fused MLP forward (1 kernel, saves activations)
+    fused loss fwd+bwd (1 kernel -> dmean/dlogits [+dlog_std] + loss)
+    fused MLP backward (L merged dgrad/wgrad kernels + 1 reduction)
+    fused multi-tensor Adam (1 kernel)
+    [PPO] fused forward + approx-KL kernel for the early-stop test
+~9 kernels per policy iteration instead of the ~40 torch-eager ones,
+with identical semantics to the eager path (same update order, same
+KL early stop, torch-exact min/clamp tie gradients — see
+loss_kernels.hip).  The DP gradient all-reduce hook runs between
+backward and step exactly as in the eager path.
+
+Falls back silently (returns False from `supported`) for policies that
+are not MLP-backed Gaussian/Categorical or are not on GPU.
+"""
+from __future__ import annotations
+
+import logging
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+from torch import Tensor
+
+from rl_replicas_amd import ops
+from rl_replicas_amd.ops.fused_mlp import _extract_layers
+from rl_replicas_amd.policies import CategoricalPolicy, GaussianPolicy
+
+logger = logging.getLogger(__name__)
+
+MODE_VPG = 0
+MODE_PPO = 1
+
+
+def _policy_kind(policy) -> Optional[str]:
+    if isinstance(policy, GaussianPolicy):
+        return "gaussian"
+    if isinstance(policy, CategoricalPolicy):
+        return "categorical"
+    return None
+
+
+def _mlp_of(module):
+    from rl_replicas_amd.networks import MLP
+
+    return module.network if isinstance(module.network, MLP) else None
+
+
+def supported(policy, obs: Tensor) -> bool:
+    if not (obs.is_cuda and ops.hip_available()):
+        return False
+    kind = _policy_kind(policy)
+    if kind is None:
+        return False
+    mlp = _mlp_of(policy)
+    if mlp is None or _extract_layers(mlp) is None:
+        return False
+    if kind == "gaussian" and policy.log_std.numel() > 32:
+        return False
+    return True
+
+
+def _forward_saved(mlp, obs: Tensor):
+    """Fused forward that saves hidden activations for backward."""
+    ext = ops._load_extension()
+    weights, biases, acts = _extract_layers(mlp)
+    outs = ext.mlp_forward(obs, list(weights), list(biases), acts, True)
+    return outs[0], outs[1:], weights, biases, acts
+
+
+def _forward_only(mlp, obs: Tensor) -> Tensor:
+    ext = ops._load_extension()
+    weights, biases, acts = _extract_layers(mlp)
+    return ext.mlp_forward(obs, list(weights), list(biases), acts, False)[0]
+
+
+def _backward_and_step(policy, mlp, obs, grad_out, hidden, final_out, weights,
+                       biases, acts, extra_grads, all_reduce_hook) -> None:
+    ext = ops._load_extension()
+    grads = ext.mlp_backward(grad_out, obs, list(weights), list(biases),
+                             list(hidden), final_out, acts)
+    n = len(weights)
+    for w, dw in zip(weights, grads[1 : 1 + n]):
+        w.grad = dw
+    for b, db in zip(biases, grads[1 + n :]):
+        b.grad = db
+    for param, grad in extra_grads:
+        param.grad = grad
+    all_reduce_hook(policy)
+    policy.optimizer.step()
+
+
+def _old_logp(policy, kind: str, obs: Tensor, actions: Tensor) -> Tensor:
+    ext = ops._load_extension()
+    mlp = _mlp_of(policy)
+    net_out = _forward_only(mlp, obs)
+    if kind == "gaussian":
+        return ext.gaussian_logp(net_out, actions, policy.log_std.data)
+    return ext.categorical_logp(net_out, actions)
+
+
+def ppo_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[str, float]:
+    """The reference PPO policy-update loop (ppo.py:173-183) on the
+    fused kernel path; returns the same metric dict."""
+    ext = ops._load_extension()
+    policy = algo.policy
+    kind = _policy_kind(policy)
+    mlp = _mlp_of(policy)
+    obs = obs.contiguous()
+    if kind == "gaussian":
+        actions_k = actions.contiguous().view(obs.shape[0], -1)
+    else:
+        actions_k = actions.contiguous().view(-1)
+    advantages = advantages.contiguous()
+
+    diagnostics = algo._policy_diagnostics(obs, actions)
+
+    with torch.no_grad():
+        old_logp = _old_logp(algo.old_policy, kind, obs, actions_k)
+
+    clip = float(algo.clip_range)
+    loss_before: Optional[Tensor] = None
+    approximate_kl = torch.zeros((), device=obs.device)
+    for i in range(algo.num_policy_gradients):
+        out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
+        if kind == "gaussian":
+            dmean, dlog_std, scalars = ext.gaussian_policy_loss(
+                out, actions_k, old_logp, advantages, policy.log_std.data, clip, MODE_PPO
+            )
+            extra = [(policy.log_std, dlog_std)]
+            grad_out = dmean
+        else:
+            dlogits, scalars = ext.categorical_policy_loss(
+                out, actions_k, old_logp, advantages, clip, MODE_PPO
+            )
+            extra = []
+            grad_out = dlogits
+        if loss_before is None:
+            loss_before = scalars[0]
+        _backward_and_step(
+            policy, mlp, obs, grad_out, hidden, out, weights, biases, acts,
+            extra, algo._all_reduce_gradients,
+        )
+        # post-step approximate KL (reference ppo.py:176-181)
+        new_out = _forward_only(mlp, obs)
+        if kind == "gaussian":
+            kl = ext.gaussian_kl(new_out, actions_k, policy.log_std.data, old_logp)
+        else:
+            kl = ext.categorical_kl(new_out, actions_k, old_logp)
+        approximate_kl = algo._reduce_scalar_mean(kl[0])
+        if float(approximate_kl) > 1.5 * algo.max_kl_divergence:
+            logger.info("Early stopping at update %d due to reaching max KL divergence.", i)
+            break
+
+    algo.old_policy.load_state_dict(policy.state_dict())
+    return {
+        "policy/loss": float(loss_before),
+        **diagnostics,
+        "policy/kl_divergence": float(approximate_kl),
+    }
+
+
+def vpg_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[str, float]:
+    """Single VPG policy step (vpg.py:194-207) on the fused path."""
+    ext = ops._load_extension()
+    policy = algo.policy
+    kind = _policy_kind(policy)
+    mlp = _mlp_of(policy)
+    obs = obs.contiguous()
+    if kind == "gaussian":
+        actions_k = actions.contiguous().view(obs.shape[0], -1)
+    else:
+        actions_k = actions.contiguous().view(-1)
+    advantages = advantages.contiguous()
+
+    diagnostics = algo._policy_diagnostics(obs, actions)
+
+    out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
+    dummy = torch.empty(0, device=obs.device)
+    if kind == "gaussian":
+        dmean, dlog_std, scalars = ext.gaussian_policy_loss(
+            out, actions_k, dummy, advantages, policy.log_std.data, 0.0, MODE_VPG
+        )
+        extra = [(policy.log_std, dlog_std)]
+        grad_out = dmean
+    else:
+        dlogits, scalars = ext.categorical_policy_loss(
+            out, actions_k, dummy, advantages, 0.0, MODE_VPG
+        )
+        extra = []
+        grad_out = dlogits
+    _backward_and_step(
+        policy, mlp, obs, grad_out, hidden, out, weights, biases, acts, extra,
+        algo._all_reduce_gradients,
+    )
+    return {"policy/loss": float(scalars[0]), **diagnostics}
+
+
+def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
+    """num_iters fused value MSE steps; returns the mean loss."""
+    ext = ops._load_extension()
+    vf = algo.value_function
+    mlp = vf.network
+    obs = obs.contiguous()
+    returns = returns.contiguous()
+    losses: List[Tensor] = []
+    for _ in range(num_iters):
+        out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
+        dv, scalars = ext.value_mse_loss(out.view(-1), returns)
+        losses.append(scalars)
+        _backward_and_step(
+            vf, mlp, obs, dv.view(out.shape), hidden, out, weights, biases, acts,
+            [], algo._all_reduce_gradients,
+        )
+    return float(torch.cat(losses).mean())
+
+
+def value_supported(algo, obs: Tensor) -> bool:
+    from rl_replicas_amd.networks import MLP
+
+    vf = algo.value_function
+    return (
+        obs.is_cuda
+        and ops.hip_available()
+        and isinstance(vf.network, MLP)
+        and _extract_layers(vf.network) is not None
+    )

@@ -22,6 +22,36 @@ __global__ void mlp_wgrad_partial_f32(const float* dy, const float* y,
                                       int batch, int out_d, int in_d, int act);
 __global__ void mlp_grad_reduce_f32(const float* workspace, float* dw, float* db,
                                     int n_blocks, int out_d, int in_d);
+__global__ void mlp_bwd_layer_f32(const float* dy, const float* y,
+                                  const float* xin, const float* W, float* dx,
+                                  float* workspace, int batch, int out_d,
+                                  int in_d, int act);
+__global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a);
+__global__ void gaussian_policy_loss_bwd(const float* mean, const float* actions,
+                                         const float* old_logp, const float* adv,
+                                         const float* log_std, float* dmean,
+                                         float* dlog_std, float* coef_ws,
+                                         float* scalars, int B, int D, float clip,
+                                         int mode);
+__global__ void gaussian_logp_kernel(const float* mean, const float* actions,
+                                     const float* log_std, float* logp, int B,
+                                     int D);
+__global__ void gaussian_kl_kernel(const float* mean, const float* actions,
+                                   const float* log_std, const float* old_logp,
+                                   float* out, int B, int D);
+__global__ void categorical_policy_loss_bwd(const float* logits,
+                                            const float* actions,
+                                            const float* old_logp,
+                                            const float* adv, float* dlogits,
+                                            float* scalars, int B, int N,
+                                            float clip, int mode);
+__global__ void categorical_logp_kernel(const float* logits, const float* actions,
+                                        float* logp, int B, int N);
+__global__ void categorical_kl_kernel(const float* logits, const float* actions,
+                                      const float* old_logp, float* out, int B,
+                                      int N);
+__global__ void value_mse_bwd_kernel(const float* v, const float* ret, float* dv,
+                                     float* scalars, int B);
 __global__ void segmented_gae_kernel(const float* rewards, const float* values,
                                      const float* last_values, const int* offsets,
                                      const int* dones, float* advantages,
@@ -119,6 +149,19 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
   auto opts = x.options();
   auto stream = current_stream();
 
+  // one workspace for every layer's per-row-block partials
+  std::vector<int64_t> totals(L), ws_off(L);
+  int64_t ws_elems = 0;
+  for (int l = 0; l < L; ++l) {
+    const int out_d = (int)weights[l].size(0);
+    const int in_d = (int)weights[l].size(1);
+    totals[l] = (int64_t)out_d * in_d + out_d;
+    ws_off[l] = ws_elems;
+    ws_elems += (int64_t)n_blocks * totals[l];
+  }
+  torch::Tensor ws = torch::empty({ws_elems}, opts);
+  float* ws_ptr = ws.data_ptr<float>();
+
   std::vector<torch::Tensor> dws(L), dbs(L);
   torch::Tensor dy = grad_out.contiguous();
   torch::Tensor dx;
@@ -127,39 +170,156 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     const int in_d = (int)weights[l].size(1);
     torch::Tensor y = (l == L - 1) ? final_out : hidden[l];
     torch::Tensor xin = (l == 0) ? x : hidden[l - 1];
-
-    // wgrad + bias grad via deterministic split-K workspace
-    torch::Tensor ws = torch::empty({(int64_t)n_blocks, (int64_t)out_d * in_d + out_d}, opts);
-    hipLaunchKernelGGL(mlp_wgrad_partial_f32, dim3(n_blocks), dim3(256), 0, stream,
-                       dy.data_ptr<float>(), y.data_ptr<float>(),
-                       xin.data_ptr<float>(), ws.data_ptr<float>(), batch, out_d,
-                       in_d, (int)acts[l]);
-    HIP_OK(hipGetLastError());
     dws[l] = torch::empty({out_d, in_d}, opts);
     dbs[l] = torch::empty({out_d}, opts);
-    int total = out_d * in_d + out_d;
-    int rb = std::min(256, (total + 255) / 256);
-    hipLaunchKernelGGL(mlp_grad_reduce_f32, dim3(rb), dim3(256), 0, stream,
-                       ws.data_ptr<float>(), dws[l].data_ptr<float>(),
-                       dbs[l].data_ptr<float>(), n_blocks, out_d, in_d);
-    HIP_OK(hipGetLastError());
-
-    // dgrad (input gradient) — needed for every layer incl. the first
-    // (the Function returns dx; unused grads are dropped by autograd)
     dx = torch::empty({batch, in_d}, opts);
-    hipLaunchKernelGGL(mlp_dgrad_f32, dim3(n_blocks), dim3(256), 0, stream,
+    // merged dgrad + wgrad/bias partials in one kernel
+    hipLaunchKernelGGL(mlp_bwd_layer_f32, dim3(n_blocks), dim3(256), 0, stream,
                        dy.data_ptr<float>(), y.data_ptr<float>(),
-                       weights[l].data_ptr<float>(), dx.data_ptr<float>(), batch,
-                       out_d, in_d, (int)acts[l]);
+                       xin.data_ptr<float>(), weights[l].data_ptr<float>(),
+                       dx.data_ptr<float>(), ws_ptr + ws_off[l], batch, out_d,
+                       in_d, (int)acts[l]);
     HIP_OK(hipGetLastError());
     dy = dx;
   }
+
+  // single deterministic reduction launch over every layer's partials
+  ReduceAllArgs ra{};
+  ra.n_layers = L;
+  ra.n_blocks = n_blocks;
+  int64_t grand = 0;
+  for (int l = 0; l < L; ++l) {
+    ra.ws[l] = ws_ptr + ws_off[l];
+    ra.dw[l] = dws[l].data_ptr<float>();
+    ra.db[l] = dbs[l].data_ptr<float>();
+    ra.total[l] = (int)totals[l];
+    ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+    grand += totals[l];
+  }
+  int rb = (int)std::min<int64_t>(512, (grand + 255) / 256);
+  hipLaunchKernelGGL(mlp_grad_reduce_all_f32, dim3(rb), dim3(256), 0, stream, ra);
+  HIP_OK(hipGetLastError());
 
   std::vector<torch::Tensor> out;
   out.push_back(dx);
   for (int l = 0; l < L; ++l) out.push_back(dws[l]);
   for (int l = 0; l < L; ++l) out.push_back(dbs[l]);
   return out;
+}
+
+// ---------------------------------------------------------------------------
+// fused losses (loss_kernels.hip)
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> gaussian_policy_loss(torch::Tensor mean,
+                                                torch::Tensor actions,
+                                                torch::Tensor old_logp,
+                                                torch::Tensor adv,
+                                                torch::Tensor log_std,
+                                                double clip, int64_t mode) {
+  check_f32_gpu(mean, "mean");
+  const int B = (int)mean.size(0);
+  const int D = (int)mean.size(1);
+  TORCH_CHECK(D <= 32, "action dim > 32 unsupported by fused loss");
+  auto opts = mean.options();
+  auto dmean = torch::empty_like(mean);
+  auto dlog_std = torch::empty({D}, opts);
+  auto coef = torch::empty({B}, opts);
+  auto scalars = torch::empty({1}, opts);
+  const float* olp = mode == 1 ? old_logp.data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL(gaussian_policy_loss_bwd, dim3(1), dim3(1024), 0,
+                     current_stream(), mean.data_ptr<float>(),
+                     actions.data_ptr<float>(), olp, adv.data_ptr<float>(),
+                     log_std.data_ptr<float>(), dmean.data_ptr<float>(),
+                     dlog_std.data_ptr<float>(), coef.data_ptr<float>(),
+                     scalars.data_ptr<float>(), B, D, (float)clip, (int)mode);
+  HIP_OK(hipGetLastError());
+  return {dmean, dlog_std, scalars};
+}
+
+torch::Tensor gaussian_logp(torch::Tensor mean, torch::Tensor actions,
+                            torch::Tensor log_std) {
+  check_f32_gpu(mean, "mean");
+  const int B = (int)mean.size(0);
+  const int D = (int)mean.size(1);
+  auto logp = torch::empty({B}, mean.options());
+  hipLaunchKernelGGL(gaussian_logp_kernel, dim3(std::min(64, (B + 1023) / 1024)),
+                     dim3(1024), 0, current_stream(), mean.data_ptr<float>(),
+                     actions.data_ptr<float>(), log_std.data_ptr<float>(),
+                     logp.data_ptr<float>(), B, D);
+  HIP_OK(hipGetLastError());
+  return logp;
+}
+
+torch::Tensor gaussian_kl(torch::Tensor mean, torch::Tensor actions,
+                          torch::Tensor log_std, torch::Tensor old_logp) {
+  check_f32_gpu(mean, "mean");
+  const int B = (int)mean.size(0);
+  const int D = (int)mean.size(1);
+  auto out = torch::empty({1}, mean.options());
+  hipLaunchKernelGGL(gaussian_kl_kernel, dim3(1), dim3(1024), 0, current_stream(),
+                     mean.data_ptr<float>(), actions.data_ptr<float>(),
+                     log_std.data_ptr<float>(), old_logp.data_ptr<float>(),
+                     out.data_ptr<float>(), B, D);
+  HIP_OK(hipGetLastError());
+  return out;
+}
+
+std::vector<torch::Tensor> categorical_policy_loss(torch::Tensor logits,
+                                                   torch::Tensor actions,
+                                                   torch::Tensor old_logp,
+                                                   torch::Tensor adv, double clip,
+                                                   int64_t mode) {
+  check_f32_gpu(logits, "logits");
+  const int B = (int)logits.size(0);
+  const int N = (int)logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  auto scalars = torch::empty({1}, logits.options());
+  const float* olp = mode == 1 ? old_logp.data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL(categorical_policy_loss_bwd, dim3(1), dim3(1024), 0,
+                     current_stream(), logits.data_ptr<float>(),
+                     actions.data_ptr<float>(), olp, adv.data_ptr<float>(),
+                     dlogits.data_ptr<float>(), scalars.data_ptr<float>(), B, N,
+                     (float)clip, (int)mode);
+  HIP_OK(hipGetLastError());
+  return {dlogits, scalars};
+}
+
+torch::Tensor categorical_logp(torch::Tensor logits, torch::Tensor actions) {
+  check_f32_gpu(logits, "logits");
+  const int B = (int)logits.size(0);
+  const int N = (int)logits.size(1);
+  auto logp = torch::empty({B}, logits.options());
+  hipLaunchKernelGGL(categorical_logp_kernel, dim3(std::min(64, (B + 1023) / 1024)),
+                     dim3(1024), 0, current_stream(), logits.data_ptr<float>(),
+                     actions.data_ptr<float>(), logp.data_ptr<float>(), B, N);
+  HIP_OK(hipGetLastError());
+  return logp;
+}
+
+torch::Tensor categorical_kl(torch::Tensor logits, torch::Tensor actions,
+                             torch::Tensor old_logp) {
+  check_f32_gpu(logits, "logits");
+  const int B = (int)logits.size(0);
+  const int N = (int)logits.size(1);
+  auto out = torch::empty({1}, logits.options());
+  hipLaunchKernelGGL(categorical_kl_kernel, dim3(1), dim3(1024), 0,
+                     current_stream(), logits.data_ptr<float>(),
+                     actions.data_ptr<float>(), old_logp.data_ptr<float>(),
+                     out.data_ptr<float>(), B, N);
+  HIP_OK(hipGetLastError());
+  return out;
+}
+
+std::vector<torch::Tensor> value_mse_loss(torch::Tensor v, torch::Tensor ret) {
+  check_f32_gpu(v, "v");
+  const int B = (int)v.numel();
+  auto dv = torch::empty_like(v);
+  auto scalars = torch::empty({1}, v.options());
+  hipLaunchKernelGGL(value_mse_bwd_kernel, dim3(1), dim3(1024), 0,
+                     current_stream(), v.data_ptr<float>(), ret.data_ptr<float>(),
+                     dv.data_ptr<float>(), scalars.data_ptr<float>(), B);
+  HIP_OK(hipGetLastError());
+  return {dv, scalars};
 }
 
 // ---------------------------------------------------------------------------
@@ -269,4 +429,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");
   m.def("fused_adam_", &fused_adam_, "fused multi-tensor Adam (gfx950)");
   m.def("fused_polyak_", &fused_polyak_, "fused multi-tensor Polyak (gfx950)");
+  m.def("gaussian_policy_loss", &gaussian_policy_loss,
+        "fused Gaussian VPG/PPO loss fwd+bwd (gfx950)");
+  m.def("gaussian_logp", &gaussian_logp, "Gaussian log-prob (gfx950)");
+  m.def("gaussian_kl", &gaussian_kl, "approx KL for Gaussian policy (gfx950)");
+  m.def("categorical_policy_loss", &categorical_policy_loss,
+        "fused Categorical VPG/PPO loss fwd+bwd (gfx950)");
+  m.def("categorical_logp", &categorical_logp, "Categorical log-prob (gfx950)");
+  m.def("categorical_kl", &categorical_kl, "approx KL for Categorical policy (gfx950)");
+  m.def("value_mse_loss", &value_mse_loss, "fused value MSE fwd+bwd (gfx950)");
 }

@@ -56,6 +56,17 @@ struct MLPArgs {
   int batch;
 };
 
+// all-layer gradient-reduction argument block (mlp_kernels.hip)
+struct ReduceAllArgs {
+  const float* ws[MLP_MAX_LAYERS];
+  float* dw[MLP_MAX_LAYERS];
+  float* db[MLP_MAX_LAYERS];
+  int total[MLP_MAX_LAYERS];  // od*id + od per layer
+  int wsize[MLP_MAX_LAYERS];  // od*id (split point)
+  int n_layers;
+  int n_blocks;
+};
+
 // multi-tensor update argument blocks (update_kernels.hip / bindings.hip)
 #define MT_MAX_TENSORS 16
 

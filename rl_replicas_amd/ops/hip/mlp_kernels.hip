@@ -245,3 +245,124 @@ __global__ void mlp_grad_reduce_f32(const float* __restrict__ workspace,
     else db[idx - out_d * in_d] = s;
   }
 }
+
+// ---------------------------------------------------------------------------
+// merged backward layer: ONE kernel per layer computing
+//   dZ = dY * act'(y)           (staged once in LDS)
+//   dX = dZ @ W                 (row tiles, written to global)
+//   dW_p = dZ^T @ X, db_p       (per-row-block partials to workspace)
+// halves the per-layer staging + launch count of the split kernels.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256, 1) void mlp_bwd_layer_f32(
+    const float* __restrict__ dy, const float* __restrict__ y,
+    const float* __restrict__ xin, const float* __restrict__ W,
+    float* __restrict__ dx, float* __restrict__ workspace, int batch,
+    int out_d, int in_d, int act) {
+  __shared__ float dz[MLP_ROWS * MLP_LDSW];
+  __shared__ float xt[MLP_ROWS * MLP_LDSW];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = blockIdx.x * MLP_ROWS;
+  const int wr0 = wave * 16;
+  float* wsp = workspace + (long)blockIdx.x * (out_d * in_d + out_d);
+
+  for (int idx = tid; idx < MLP_ROWS * out_d; idx += 256) {
+    int r = idx / out_d, c = idx % out_d;
+    int row = row0 + r;
+    float v = 0.f;
+    if (row < batch) {
+      long g = (long)row * out_d + c;
+      v = dy[g] * act_grad_from_y(act, y[g]);
+    }
+    dz[r * MLP_LDSW + c] = v;
+  }
+  load_tile_f32(xin, xt, row0, batch, in_d, tid);
+  __syncthreads();
+
+  const int i = lane & 15;
+  const int k = lane >> 4;
+
+  // ---- dgrad: dX[b][j] = sum_k dZ[b][k] W[k][j] ----
+  for (int jt = 0; jt < in_d; jt += 16) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int j = jt + i;
+    const bool jok = j < in_d;
+    for (int k0 = 0; k0 < out_d; k0 += 4) {
+      const int kk = k0 + k;
+      float a = (kk < out_d) ? dz[(wr0 + i) * MLP_LDSW + kk] : 0.f;
+      float bv = (jok && kk < out_d) ? W[(long)kk * in_d + j] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    }
+    if (jok) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + wr0 + (lane >> 4) * 4 + r;
+        if (row < batch) dx[(long)row * in_d + jt + i] = acc[r];
+      }
+    }
+  }
+
+  // ---- wgrad partials: dW[i][j] = sum_r dZ[r][i] X[r][j] ----
+  const int n_it = (out_d + 15) / 16;
+  for (int it = wave; it < n_it; it += 4) {
+    const int ii = it * 16 + i;
+    for (int jt = 0; jt < in_d; jt += 16) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int k0 = 0; k0 < MLP_ROWS; k0 += 4) {
+        float a = (ii < out_d) ? dz[(k0 + k) * MLP_LDSW + ii] : 0.f;
+        float bv = (jt + i < in_d) ? xt[(k0 + k) * MLP_LDSW + jt + i] : 0.f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+      }
+      const int col = jt + i;
+      if (col < in_d) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int orow = it * 16 + (lane >> 4) * 4 + r;
+          if (orow < out_d) wsp[(long)orow * in_d + col] = acc[r];
+        }
+      }
+    }
+  }
+
+  __syncthreads();
+  for (int c = tid; c < out_d; c += 256) {
+    float s = 0.f;
+    #pragma unroll 4
+    for (int r = 0; r < MLP_ROWS; ++r) s += dz[r * MLP_LDSW + c];
+    wsp[(long)out_d * in_d + c] = s;
+  }
+}
+
+// all-layer deterministic partial reduction: one launch per backward.
+// workspace holds per-layer segments of n_blocks partials each;
+// fixed (s0+s1)+(s2+s3) accumulation order -> bitwise reproducible.
+__global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a) {
+  // flatten all layers' elements into one grid-stride loop
+  int grand = 0;
+  int base[MLP_MAX_LAYERS];
+  for (int l = 0; l < a.n_layers; ++l) {
+    base[l] = grand;
+    grand += a.total[l];
+  }
+  for (int g = blockIdx.x * blockDim.x + threadIdx.x; g < grand;
+       g += gridDim.x * blockDim.x) {
+    int l = 0;
+    while (l + 1 < a.n_layers && g >= base[l + 1]) ++l;
+    const int idx = g - base[l];
+    const long stride = a.total[l];
+    const float* ws = a.ws[l];
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    int p = 0;
+    for (; p + 3 < a.n_blocks; p += 4) {
+      s0 += ws[(p + 0) * stride + idx];
+      s1 += ws[(p + 1) * stride + idx];
+      s2 += ws[(p + 2) * stride + idx];
+      s3 += ws[(p + 3) * stride + idx];
+    }
+    for (; p < a.n_blocks; ++p) s0 += ws[p * stride + idx];
+    const float s = (s0 + s1) + (s2 + s3);
+    if (idx < a.wsize[l]) a.dw[l][idx] = s;
+    else a.db[l][idx - a.wsize[l]] = s;
+  }
+}

@@ -22,6 +22,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(HIP_DIR, "bindings.hip"),
         os.path.join(HIP_DIR, "mlp_kernels.hip"),
+        os.path.join(HIP_DIR, "loss_kernels.hip"),
         os.path.join(HIP_DIR, "rollout_kernels.hip"),
         os.path.join(HIP_DIR, "update_kernels.hip"),
     ],
