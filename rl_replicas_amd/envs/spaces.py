@@ -66,19 +66,22 @@ class Box(Space):
         self.bounded_below = np.isfinite(self.low)
         self.bounded_above = np.isfinite(self.high)
 
-    def sample(self) -> np.ndarray:
+    def sample(self, batch: Optional[int] = None) -> np.ndarray:
         # Bounded dims: uniform in [low, high); unbounded: standard normal
         # (matches gymnasium.Box.sample semantics closely enough for
         # exploration warm-up; the reference only uses bounded Boxes).
-        out = np.empty(self.shape, dtype=np.float64)
-        bounded = self.bounded_below & self.bounded_above
-        u = self.np_random.random(self.shape)
+        # `batch` (extension): draw a vectorized [batch, *shape] sample in
+        # one call — used by RandomPolicy over vectorized envs.
+        shape = self.shape if batch is None else (batch,) + self.shape
+        out = np.empty(shape, dtype=np.float64)
+        bounded = np.broadcast_to(self.bounded_below & self.bounded_above, shape)
+        u = self.np_random.random(shape)
         out[bounded] = (self.low + u * (self.high - self.low))[bounded]
         if not bounded.all():
-            n = self.np_random.standard_normal(self.shape)
-            only_below = self.bounded_below & ~self.bounded_above
-            only_above = ~self.bounded_below & self.bounded_above
-            neither = ~self.bounded_below & ~self.bounded_above
+            n = self.np_random.standard_normal(shape)
+            only_below = np.broadcast_to(self.bounded_below & ~self.bounded_above, shape)
+            only_above = np.broadcast_to(~self.bounded_below & self.bounded_above, shape)
+            neither = np.broadcast_to(~self.bounded_below & ~self.bounded_above, shape)
             out[only_below] = (self.low + np.abs(n))[only_below]
             out[only_above] = (self.high - np.abs(n))[only_above]
             out[neither] = n[neither]
@@ -100,7 +103,9 @@ class Discrete(Space):
         self.n = int(n)
         self.start = int(start)
 
-    def sample(self) -> np.int64:
+    def sample(self, batch: Optional[int] = None):
+        if batch is not None:
+            return self.start + self.np_random.integers(self.n, size=batch)
         return np.int64(self.start + self.np_random.integers(self.n))
 
     def contains(self, x) -> bool:

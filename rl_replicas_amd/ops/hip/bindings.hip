@@ -14,18 +14,12 @@
 #include "common.h"
 
 // kernel declarations (defined in the .hip translation units)
-__global__ void fused_mlp_fwd_f32(MLPArgs args, const float* x, int save_hidden);
-__global__ void mlp_dgrad_f32(const float* dy, const float* y, const float* W,
-                              float* dx, int batch, int out_d, int in_d, int act);
-__global__ void mlp_wgrad_partial_f32(const float* dy, const float* y,
-                                      const float* xin, float* workspace,
-                                      int batch, int out_d, int in_d, int act);
-__global__ void mlp_grad_reduce_f32(const float* workspace, float* dw, float* db,
-                                    int n_blocks, int out_d, int in_d);
-__global__ void mlp_bwd_layer_f32(const float* dy, const float* y,
-                                  const float* xin, const float* W, float* dx,
-                                  float* workspace, int batch, int out_d,
-                                  int in_d, int act);
+void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
+                    int rows, int maxw, int n_blocks, hipStream_t stream);
+void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
+                          const float* W, float* dx, float* ws, int batch,
+                          int out_d, int in_d, int act, int rows, int maxw,
+                          int n_blocks, hipStream_t stream);
 __global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a);
 __global__ void gaussian_policy_loss_bwd(const float* mean, const float* actions,
                                          const float* old_logp, const float* adv,
@@ -52,6 +46,13 @@ __global__ void categorical_kl_kernel(const float* logits, const float* actions,
                                       int N);
 __global__ void value_mse_bwd_kernel(const float* v, const float* ret, float* dv,
                                      float* scalars, int B);
+__global__ void gaussian_sample_kernel(const float* mean, const float* log_std,
+                                       float* out, int B, int D, uint64_t seed,
+                                       uint64_t offset, float noise_scale,
+                                       float limit);
+__global__ void categorical_sample_kernel(const float* logits, int64_t* out,
+                                          int B, int N, uint64_t seed,
+                                          uint64_t offset);
 __global__ void segmented_gae_kernel(const float* rewards, const float* values,
                                      const float* last_values, const int* offsets,
                                      const int* dones, float* advantages,
@@ -78,6 +79,21 @@ void check_f32_gpu(const torch::Tensor& t, const char* name) {
     hipError_t _e = (expr);                                                 \
     TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));    \
   } while (0)
+
+// (ROWS, MAXW) tile selection for the templated MLP kernels.
+// MAXW: smallest instantiated width bound that fits every layer ->
+// smaller LDS footprint, higher per-CU occupancy for the narrow
+// on-policy nets.  ROWS: 32 at large batches fills the 256-CU chip
+// (4000-row batch -> 125 workgroups); 64 otherwise.
+void pick_tile(int batch, int max_width, int* rows, int* maxw) {
+  *maxw = max_width <= 64 ? 64 : 256;
+  *rows = batch >= 2048 ? 32 : 64;
+  static int env_rows = []() {
+    const char* e = getenv("RL_REPLICAS_AMD_MLP_ROWS");
+    return e ? atoi(e) : 0;
+  }();
+  if (env_rows == 32 || env_rows == 64) *rows = env_rows;
+}
 
 }  // namespace
 
@@ -125,11 +141,14 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
   }
   args.h[L - 1] = final_out.data_ptr<float>();
 
-  const int n_blocks = (args.batch + MLP_ROWS - 1) / MLP_ROWS;
+  int max_width = args.dims[0];
+  for (int l = 1; l <= L; ++l) max_width = std::max(max_width, args.dims[l]);
+  int rows, maxw;
+  pick_tile(args.batch, max_width, &rows, &maxw);
+  const int n_blocks = (args.batch + rows - 1) / rows;
   if (n_blocks > 0) {
-    hipLaunchKernelGGL(fused_mlp_fwd_f32, dim3(n_blocks), dim3(256), 0,
-                       current_stream(), args, x.data_ptr<float>(),
-                       save_hidden ? 1 : 0);
+    launch_mlp_fwd(args, x.data_ptr<float>(), save_hidden ? 1 : 0, rows, maxw,
+                   n_blocks, current_stream());
     HIP_OK(hipGetLastError());
   }
   return outs;
@@ -145,7 +164,12 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
   check_f32_gpu(grad_out, "grad_out");
   check_f32_gpu(x, "x");
   const int batch = (int)x.size(0);
-  const int n_blocks = (batch + MLP_ROWS - 1) / MLP_ROWS;
+  int max_width = (int)x.size(1);
+  for (int l = 0; l < L; ++l)
+    max_width = std::max(max_width, (int)weights[l].size(0));
+  int rows, maxw;
+  pick_tile(batch, max_width, &rows, &maxw);
+  const int n_blocks = (batch + rows - 1) / rows;
   auto opts = x.options();
   auto stream = current_stream();
 
@@ -174,11 +198,10 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dbs[l] = torch::empty({out_d}, opts);
     dx = torch::empty({batch, in_d}, opts);
     // merged dgrad + wgrad/bias partials in one kernel
-    hipLaunchKernelGGL(mlp_bwd_layer_f32, dim3(n_blocks), dim3(256), 0, stream,
-                       dy.data_ptr<float>(), y.data_ptr<float>(),
-                       xin.data_ptr<float>(), weights[l].data_ptr<float>(),
-                       dx.data_ptr<float>(), ws_ptr + ws_off[l], batch, out_d,
-                       in_d, (int)acts[l]);
+    launch_mlp_bwd_layer(dy.data_ptr<float>(), y.data_ptr<float>(),
+                         xin.data_ptr<float>(), weights[l].data_ptr<float>(),
+                         dx.data_ptr<float>(), ws_ptr + ws_off[l], batch, out_d,
+                         in_d, (int)acts[l], rows, maxw, n_blocks, stream);
     HIP_OK(hipGetLastError());
     dy = dx;
   }
@@ -421,6 +444,39 @@ void fused_polyak_(std::vector<torch::Tensor> srcs, std::vector<torch::Tensor> d
   }
 }
 
+torch::Tensor gaussian_sample(torch::Tensor mean, torch::Tensor log_std,
+                              int64_t seed, int64_t offset, double noise_scale,
+                              double limit) {
+  check_f32_gpu(mean, "mean");
+  const int B = (int)mean.size(0);
+  const int D = (int)mean.size(1);
+  auto out = torch::empty_like(mean);
+  const int total = B * D;
+  hipLaunchKernelGGL(gaussian_sample_kernel,
+                     dim3(std::min(256, (total + 255) / 256)), dim3(256), 0,
+                     current_stream(), mean.data_ptr<float>(),
+                     log_std.data_ptr<float>(), out.data_ptr<float>(), B, D,
+                     (uint64_t)seed, (uint64_t)offset, (float)noise_scale,
+                     (float)limit);
+  HIP_OK(hipGetLastError());
+  return out;
+}
+
+torch::Tensor categorical_sample(torch::Tensor logits, int64_t seed,
+                                 int64_t offset) {
+  check_f32_gpu(logits, "logits");
+  const int B = (int)logits.size(0);
+  const int N = (int)logits.size(1);
+  auto out = torch::empty({B}, logits.options().dtype(torch::kInt64));
+  hipLaunchKernelGGL(categorical_sample_kernel,
+                     dim3(std::min(256, (B + 255) / 256)), dim3(256), 0,
+                     current_stream(), logits.data_ptr<float>(),
+                     out.data_ptr<int64_t>(), B, N, (uint64_t)seed,
+                     (uint64_t)offset);
+  HIP_OK(hipGetLastError());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlp_forward", &mlp_forward, "fused MLP forward (gfx950)");
   m.def("mlp_backward", &mlp_backward, "fused MLP backward (gfx950)");
@@ -438,4 +494,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("categorical_logp", &categorical_logp, "Categorical log-prob (gfx950)");
   m.def("categorical_kl", &categorical_kl, "approx KL for Categorical policy (gfx950)");
   m.def("value_mse_loss", &value_mse_loss, "fused value MSE fwd+bwd (gfx950)");
+  m.def("gaussian_sample", &gaussian_sample,
+        "Philox Gaussian action sample (gfx950)");
+  m.def("categorical_sample", &categorical_sample,
+        "Philox categorical action sample (gfx950)");
 }

@@ -17,7 +17,28 @@ class CategoricalPolicy(StochasticPolicy):
         super().__init__()
         self.network = network
         self.optimizer = optimizer
+        self._sample_offset = 0
 
     def forward(self, observation: Tensor) -> Categorical:
         logits: Tensor = self.network(observation)
         return Categorical(logits=logits)
+
+    def get_action_tensor(self, observation):
+        # GPU fast path: fused MLP forward + one Philox sample kernel
+        from rl_replicas_amd import ops
+        import torch
+
+        if observation.dim() == 2 and ops.wants_hip(observation):
+            from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+            if _extract_layers(self.network) is not None:
+                with torch.no_grad():
+                    logits = self.network(observation)
+                    ext = ops._load_extension()
+                    self._sample_offset += 1
+                    return ext.categorical_sample(
+                        logits,
+                        torch.initial_seed() & 0x7FFFFFFFFFFFFFFF,
+                        self._sample_offset,
+                    )
+        return super().get_action_tensor(observation)

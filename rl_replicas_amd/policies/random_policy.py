@@ -17,9 +17,14 @@ class RandomPolicy(Policy):
         self.action_space = action_space
 
     def _sample(self, observation_shape) -> np.ndarray:
-        # batched obs [B, D] -> B independent samples
+        # batched obs [B, D] -> B independent samples (one vectorized draw
+        # when the space supports it)
         if len(observation_shape) > 1:
-            return np.stack([self.action_space.sample() for _ in range(observation_shape[0])])
+            batch = observation_shape[0]
+            try:
+                return np.asarray(self.action_space.sample(batch))
+            except TypeError:  # third-party space without batch support
+                return np.stack([self.action_space.sample() for _ in range(batch)])
         return np.asarray(self.action_space.sample())
 
     def get_action_tensor(self, observation: Tensor) -> Tensor:

@@ -7,25 +7,30 @@ is ONE batched policy forward over obs[N, O] (one H2D + one fused MLP
 kernel on GPU) plus ONE vectorized env transition, so `num_samples`
 total steps take `num_samples / N` policy calls.
 
-Episode-slicing semantics match BatchSampler exactly per instance:
-episodes end on terminated|truncated (both flagged `done`, matching
-the reference's bootstrap semantics) or at epoch end (trajectory cut,
-`done=False`, bootstrapped with V(s_last)); `is_continuous=True`
-retains env state across calls; the env is seeded on first reset only.
+The per-step work is O(1) Python (append five array references);
+rollout storage is step-major, and episode slicing happens once at the
+end of `sample()` with numpy boundary searches — no per-instance Python
+loop in the hot path.  Episodes are emitted in instance-major order
+(all of instance 0's slices, then instance 1's, ...), deterministic
+given the seed, with per-episode data exposed as contiguous array
+views (compatible with the reference's List[List[...]] Experience
+protocol).
 
-Episodes are emitted in instance-major order (all of instance 0's
-slices, then instance 1's, ...) — deterministic given the seed.
+Episode semantics match BatchSampler exactly per instance: episodes end
+on terminated|truncated (both flagged `done`, matching the reference's
+bootstrap semantics) or at epoch end (trajectory cut, `done=False`,
+bootstrapped with V(s_last)); `is_continuous=True` retains env state
+across calls; the env is seeded on first reset only.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import Optional
 
 import numpy as np
 
 from rl_replicas_amd.envs.vector import VectorEnv
 from rl_replicas_amd.experience import Experience
 from rl_replicas_amd.policies import Policy
-from rl_replicas_amd.samplers.batch_sampler import _EpisodeAccumulator
 from rl_replicas_amd.samplers.sampler import Sampler
 
 
@@ -50,27 +55,63 @@ class VectorSampler(Sampler):
         elif not self.is_continuous:
             self.observations = self.env.reset()
 
-        accs = [_EpisodeAccumulator() for _ in range(n_envs)]
-        # per-instance episode slices, stitched instance-major at the end
-        slices: List[List] = [[] for _ in range(n_envs)]
-
-        for step in range(steps):
+        # step-major storage: the hot loop only appends references
+        obs_steps = []
+        act_steps = []
+        rew_steps = []
+        done_steps = []
+        final_steps = []
+        for _ in range(steps):
             obs = self.observations
             actions = np.asarray(policy.get_action_numpy(obs))
             next_obs, rewards, terminated, truncated, final_obs = self.env.step(actions)
-            done = terminated | truncated
-            epoch_ended = step == steps - 1
-
-            for i in range(n_envs):
-                accs[i].push(obs[i], actions[i], rewards[i], bool(done[i]))
-                if done[i] or epoch_ended:
-                    # final_obs holds the true successor state (pre-autoreset)
-                    slices[i].append((accs[i], np.asarray(final_obs[i])))
-                    accs[i] = _EpisodeAccumulator()
+            obs_steps.append(obs)
+            act_steps.append(actions)
+            rew_steps.append(rewards)
+            done_steps.append(terminated | truncated)
+            final_steps.append(final_obs)
             self.observations = next_obs
+
+        # [T, B, ...] views for slicing
+        obs_arr = np.stack(obs_steps)
+        act_arr = np.stack(act_steps)
+        rew_arr = np.stack(rew_steps)
+        done_arr = np.stack(done_steps)
 
         experience = Experience()
         for i in range(n_envs):
-            for acc, last_obs in slices[i]:
-                acc.flush_into(experience, last_obs)
+            done_i = done_arr[:, i]
+            boundaries = np.flatnonzero(done_i)
+            start = 0
+            for b in boundaries:
+                self._emit(
+                    experience,
+                    obs_arr[start : b + 1, i],
+                    act_arr[start : b + 1, i],
+                    rew_arr[start : b + 1, i],
+                    done_i[start : b + 1],
+                    np.asarray(final_steps[b][i]),
+                )
+                start = b + 1
+            if start < steps:
+                # epoch-end trajectory cut: bootstrap obs is the live state
+                self._emit(
+                    experience,
+                    obs_arr[start:steps, i],
+                    act_arr[start:steps, i],
+                    rew_arr[start:steps, i],
+                    done_i[start:steps],
+                    np.asarray(self.observations[i]),
+                )
         return experience
+
+    @staticmethod
+    def _emit(experience: Experience, obs, acts, rews, dones, last_obs) -> None:
+        # contiguous per-episode views satisfy the List[List[...]] protocol
+        experience.observations.append(obs)
+        experience.actions.append(acts)
+        experience.rewards.append(rews)
+        experience.dones.append(dones)
+        experience.last_observations.append(last_obs)
+        experience.episode_returns.append(float(np.sum(rews)))
+        experience.episode_lengths.append(int(len(rews)))

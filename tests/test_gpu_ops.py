@@ -219,3 +219,39 @@ class TestFusedPolyak:
         polyak_average(net.parameters(), tgt.parameters(), 0.995)
         for t, e in zip(tgt.parameters(), expected):
             torch.testing.assert_close(t.detach(), e, rtol=1e-6, atol=1e-7)
+
+
+class TestSampleKernels:
+    def test_gaussian_sample_statistics(self, ext):
+        B, D = 20000, 4
+        mean = torch.randn(1, D, device="cuda").expand(B, D).contiguous()
+        log_std = torch.tensor([-0.5, 0.0, 0.3, -1.0], device="cuda")
+        out = ext.gaussian_sample(mean, log_std, 12345, 1, -1.0, -1.0)
+        emp_mean = out.mean(0)
+        emp_std = out.std(0)
+        torch.testing.assert_close(emp_mean, mean[0], atol=0.05, rtol=0.0)
+        torch.testing.assert_close(emp_std, torch.exp(log_std), atol=0.05, rtol=0.05)
+
+    def test_gaussian_sample_deterministic_and_offset_varies(self, ext):
+        mean = torch.zeros(100, 2, device="cuda")
+        ls = torch.zeros(2, device="cuda")
+        a = ext.gaussian_sample(mean, ls, 7, 3, -1.0, -1.0)
+        b = ext.gaussian_sample(mean, ls, 7, 3, -1.0, -1.0)
+        c = ext.gaussian_sample(mean, ls, 7, 4, -1.0, -1.0)
+        assert torch.equal(a, b)
+        assert not torch.equal(a, c)
+
+    def test_gaussian_sample_clip(self, ext):
+        mean = torch.zeros(1000, 2, device="cuda")
+        ls = torch.zeros(2, device="cuda")
+        out = ext.gaussian_sample(mean, ls, 1, 1, 5.0, 2.0)  # big noise, clip 2
+        assert float(out.abs().max()) <= 2.0
+
+    def test_categorical_sample_frequencies(self, ext):
+        B = 40000
+        logits = torch.log(torch.tensor([[0.1, 0.2, 0.7]], device="cuda")).expand(B, 3).contiguous()
+        out = ext.categorical_sample(logits, 99, 1)
+        freqs = torch.bincount(out, minlength=3).float() / B
+        torch.testing.assert_close(
+            freqs, torch.tensor([0.1, 0.2, 0.7], device="cuda"), atol=0.01, rtol=0.0
+        )
