@@ -58,7 +58,7 @@ def supported(policy, obs: Tensor) -> bool:
     mlp = _mlp_of(policy)
     if mlp is None or _extract_layers(mlp) is None:
         return False
-    if kind == "gaussian" and policy.log_std.numel() > 32:
+    if kind == "gaussian" and policy.log_std.numel() > 8:
         return False
     return True
 
@@ -102,6 +102,193 @@ def _old_logp(policy, kind: str, obs: Tensor, actions: Tensor) -> Tensor:
     return ext.categorical_logp(net_out, actions)
 
 
+# ---------------------------------------------------------------------------
+# hipGraph-captured loops
+#
+# The per-iteration kernel sequences are captured once per rollout shape
+# into hipGraphs and replayed, removing every host launch from the
+# 80-iteration update loops.  The Adam step counter lives on device, so
+# replays advance optimizer state correctly; inputs (obs / actions /
+# advantages / old_logp / returns) are copied into capture-stable
+# buffers each epoch.  Graphs are used only when (a) the optimizer is
+# the FusedAdam (device-side state) and (b) the run is single-process
+# (collectives are not captured); otherwise the eager fused path runs.
+# ---------------------------------------------------------------------------
+
+
+def _graphs_common(algo) -> bool:
+    import os
+
+    if os.environ.get("RL_REPLICAS_AMD_DISABLE_GRAPHS", "0") == "1":
+        return False
+    from rl_replicas_amd.parallel.ddp import distributed_is_active
+
+    return not (distributed_is_active() or getattr(algo, "_dp_enabled", False))
+
+
+def _graphs_enabled(algo) -> bool:
+    from rl_replicas_amd.ops.fused_adam import FusedAdam
+
+    return _graphs_common(algo) and isinstance(algo.policy.optimizer, FusedAdam)
+
+
+def _graphs_enabled_value(algo) -> bool:
+    from rl_replicas_amd.ops.fused_adam import FusedAdam
+
+    return _graphs_common(algo) and isinstance(algo.value_function.optimizer, FusedAdam)
+
+
+def _ensure_adam_state(optimizer) -> List[Tensor]:
+    """Create (zero) Adam state for every param up front; return all
+    state tensors (for snapshot/restore around graph warmup)."""
+    tensors: List[Tensor] = []
+    for group in optimizer.param_groups:
+        for p in group["params"]:
+            state = optimizer.state[p]
+            if len(state) == 0:
+                state["step"] = torch.zeros((), dtype=torch.float32, device=p.device)
+                state["exp_avg"] = torch.zeros_like(p)
+                state["exp_avg_sq"] = torch.zeros_like(p)
+            tensors += [state["step"], state["exp_avg"], state["exp_avg_sq"]]
+    return tensors
+
+
+class _CapturedLoop:
+    """Capture `body()` into a hipGraph with state snapshot/restore
+    around the warmup runs (warmup executes real kernels; capture does
+    not execute)."""
+
+    def __init__(self, body, state_tensors: List[Tensor]):
+        snapshot = [t.detach().clone() for t in state_tensors]
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                body()
+        torch.cuda.current_stream().wait_stream(stream)
+        with torch.no_grad():
+            for t, snap in zip(state_tensors, snapshot):
+                t.copy_(snap)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.outputs = body()
+
+    def replay(self):
+        self.graph.replay()
+        return self.outputs
+
+
+class _GraphedPPO:
+    """One captured PPO policy iteration: fwd + loss + bwd + Adam +
+    KL-eval forward, replayed up to num_policy_gradients times."""
+
+    def __init__(self, algo, kind: str, obs0: Tensor, actions0: Tensor,
+                 adv0: Tensor, old_logp0: Tensor):
+        ext = ops._load_extension()
+        policy = algo.policy
+        mlp = _mlp_of(policy)
+        self.kind = kind
+        # capture-stable input buffers, pre-filled with real data so the
+        # warmup runs on valid values
+        self.obs = obs0.clone()
+        self.actions = actions0.clone()
+        self.adv = adv0.clone()
+        self.old_logp = old_logp0.clone()
+        clip = float(algo.clip_range)
+
+        def body():
+            out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
+            if kind == "gaussian":
+                dmean, dlog_std, scalars = ext.gaussian_policy_loss(
+                    out, self.actions, self.old_logp, self.adv,
+                    policy.log_std.data, clip, MODE_PPO,
+                )
+                extra = [(policy.log_std, dlog_std)]
+                grad_out = dmean
+            else:
+                dlogits, scalars = ext.categorical_policy_loss(
+                    out, self.actions, self.old_logp, self.adv, clip, MODE_PPO
+                )
+                extra = []
+                grad_out = dlogits
+            _backward_and_step(
+                policy, mlp, self.obs, grad_out, hidden, out, weights, biases,
+                acts, extra, lambda _m: None,
+            )
+            new_out = _forward_only(mlp, self.obs)
+            if kind == "gaussian":
+                kl = ext.gaussian_kl(new_out, self.actions, policy.log_std.data, self.old_logp)
+            else:
+                kl = ext.categorical_kl(new_out, self.actions, self.old_logp)
+            return scalars, kl
+
+        state = [p.data for p in policy.parameters()]
+        state += _ensure_adam_state(policy.optimizer)
+        self.loop = _CapturedLoop(body, state)
+
+    def run(self, algo, obs, actions, advantages, old_logp) -> Dict[str, float]:
+        self.obs.copy_(obs)
+        self.actions.copy_(actions.view(self.actions.shape))
+        self.adv.copy_(advantages)
+        self.old_logp.copy_(old_logp)
+        loss_before = None
+        approximate_kl = 0.0
+        for i in range(algo.num_policy_gradients):
+            scalars, kl = self.loop.replay()
+            if loss_before is None:
+                loss_before = float(scalars[0])
+            approximate_kl = float(kl[0])
+            if approximate_kl > 1.5 * algo.max_kl_divergence:
+                logger.info(
+                    "Early stopping at update %d due to reaching max KL divergence.", i
+                )
+                break
+        return {"policy/loss": loss_before, "policy/kl_divergence": approximate_kl}
+
+
+class _GraphedValueLoop:
+    """The whole num_value_gradients value-function loop as ONE graph."""
+
+    def __init__(self, algo, obs0: Tensor, returns0: Tensor, num_iters: int):
+        ext = ops._load_extension()
+        vf = algo.value_function
+        mlp = vf.network
+        self.obs = obs0.clone()
+        self.returns = returns0.clone()
+        self.num_iters = num_iters
+
+        def body():
+            losses = []
+            for _ in range(num_iters):
+                out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
+                dv, scalars = ext.value_mse_loss(out.view(-1), self.returns)
+                losses.append(scalars[:1])
+                _backward_and_step(
+                    vf, mlp, self.obs, dv.view(out.shape), hidden, out, weights,
+                    biases, acts, [], lambda _m: None,
+                )
+            return losses
+
+        state = [p.data for p in vf.parameters()]
+        state += _ensure_adam_state(vf.optimizer)
+        self.loop = _CapturedLoop(body, state)
+
+    def run(self, obs, returns) -> float:
+        self.obs.copy_(obs)
+        self.returns.copy_(returns)
+        losses = self.loop.replay()
+        return float(torch.cat(losses).mean())
+
+
+def _get_cached_graph(algo, attr: str, key, builder):
+    cached = getattr(algo, attr, None)
+    if cached is not None and cached[0] == key:
+        return cached[1]
+    graph = builder()
+    setattr(algo, attr, (key, graph))
+    return graph
+
+
 def ppo_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[str, float]:
     """The reference PPO policy-update loop (ppo.py:173-183) on the
     fused kernel path; returns the same metric dict."""
@@ -120,6 +307,21 @@ def ppo_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[s
 
     with torch.no_grad():
         old_logp = _old_logp(algo.old_policy, kind, obs, actions_k)
+
+    if _graphs_enabled(algo):
+        key = (kind, tuple(obs.shape), tuple(actions_k.shape),
+               tuple(id(p) for p in policy.parameters()))
+        graphed = _get_cached_graph(
+            algo, "_ppo_policy_graph", key,
+            lambda: _GraphedPPO(algo, kind, obs, actions_k, advantages, old_logp),
+        )
+        metrics = graphed.run(algo, obs, actions_k, advantages, old_logp)
+        algo.old_policy.load_state_dict(policy.state_dict())
+        return {
+            "policy/loss": metrics["policy/loss"],
+            **diagnostics,
+            "policy/kl_divergence": metrics["policy/kl_divergence"],
+        }
 
     clip = float(algo.clip_range)
     loss_before: Optional[Tensor] = None
@@ -206,11 +408,20 @@ def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
     mlp = vf.network
     obs = obs.contiguous()
     returns = returns.contiguous()
+    from rl_replicas_amd.ops.fused_adam import FusedAdam
+
+    if _graphs_enabled_value(algo):
+        key = (tuple(obs.shape), num_iters, tuple(id(p) for p in vf.parameters()))
+        graphed = _get_cached_graph(
+            algo, "_value_graph", key,
+            lambda: _GraphedValueLoop(algo, obs, returns, num_iters),
+        )
+        return graphed.run(obs, returns)
     losses: List[Tensor] = []
     for _ in range(num_iters):
         out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
         dv, scalars = ext.value_mse_loss(out.view(-1), returns)
-        losses.append(scalars)
+        losses.append(scalars[:1])
         _backward_and_step(
             vf, mlp, obs, dv.view(out.shape), hidden, out, weights, biases, acts,
             [], algo._all_reduce_gradients,

@@ -21,12 +21,13 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
                           int out_d, int in_d, int act, int rows, int maxw,
                           int n_blocks, hipStream_t stream);
 __global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a);
-__global__ void gaussian_policy_loss_bwd(const float* mean, const float* actions,
-                                         const float* old_logp, const float* adv,
-                                         const float* log_std, float* dmean,
-                                         float* dlog_std, float* coef_ws,
-                                         float* scalars, int B, int D, float clip,
-                                         int mode);
+void launch_gaussian_loss(const float* mean, const float* actions,
+                          const float* old_logp, const float* adv,
+                          const float* log_std, float* dmean, float* partials,
+                          int B, int D, float clip, int mode, int n_blocks,
+                          hipStream_t stream);
+__global__ void loss_partials_finalize(const float* partials, float* dlog_std,
+                                       float* scalars, int n_blocks, int D);
 __global__ void gaussian_logp_kernel(const float* mean, const float* actions,
                                      const float* log_std, float* logp, int B,
                                      int D);
@@ -37,7 +38,7 @@ __global__ void categorical_policy_loss_bwd(const float* logits,
                                             const float* actions,
                                             const float* old_logp,
                                             const float* adv, float* dlogits,
-                                            float* scalars, int B, int N,
+                                            float* partials, int B, int N,
                                             float clip, int mode);
 __global__ void categorical_logp_kernel(const float* logits, const float* actions,
                                         float* logp, int B, int N);
@@ -242,19 +243,23 @@ std::vector<torch::Tensor> gaussian_policy_loss(torch::Tensor mean,
   check_f32_gpu(mean, "mean");
   const int B = (int)mean.size(0);
   const int D = (int)mean.size(1);
-  TORCH_CHECK(D <= 32, "action dim > 32 unsupported by fused loss");
+  TORCH_CHECK(D >= 1 && D <= 8, "action dim must be in [1,8] for the fused loss");
   auto opts = mean.options();
   auto dmean = torch::empty_like(mean);
   auto dlog_std = torch::empty({D}, opts);
-  auto coef = torch::empty({B}, opts);
   auto scalars = torch::empty({1}, opts);
+  const int n_blocks = std::min(32, (B + 255) / 256);
+  auto partials = torch::empty({n_blocks, D + 1}, opts);
   const float* olp = mode == 1 ? old_logp.data_ptr<float>() : nullptr;
-  hipLaunchKernelGGL(gaussian_policy_loss_bwd, dim3(1), dim3(1024), 0,
-                     current_stream(), mean.data_ptr<float>(),
-                     actions.data_ptr<float>(), olp, adv.data_ptr<float>(),
-                     log_std.data_ptr<float>(), dmean.data_ptr<float>(),
-                     dlog_std.data_ptr<float>(), coef.data_ptr<float>(),
-                     scalars.data_ptr<float>(), B, D, (float)clip, (int)mode);
+  auto stream = current_stream();
+  launch_gaussian_loss(mean.data_ptr<float>(), actions.data_ptr<float>(), olp,
+                       adv.data_ptr<float>(), log_std.data_ptr<float>(),
+                       dmean.data_ptr<float>(), partials.data_ptr<float>(), B,
+                       D, (float)clip, (int)mode, n_blocks, stream);
+  HIP_OK(hipGetLastError());
+  hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(D + 1), 0, stream,
+                     partials.data_ptr<float>(), dlog_std.data_ptr<float>(),
+                     scalars.data_ptr<float>(), n_blocks, D);
   HIP_OK(hipGetLastError());
   return {dmean, dlog_std, scalars};
 }
@@ -297,12 +302,19 @@ std::vector<torch::Tensor> categorical_policy_loss(torch::Tensor logits,
   const int N = (int)logits.size(1);
   auto dlogits = torch::empty_like(logits);
   auto scalars = torch::empty({1}, logits.options());
+  const int n_blocks = std::min(32, (B + 255) / 256);
+  auto partials = torch::empty({n_blocks}, logits.options());
   const float* olp = mode == 1 ? old_logp.data_ptr<float>() : nullptr;
-  hipLaunchKernelGGL(categorical_policy_loss_bwd, dim3(1), dim3(1024), 0,
-                     current_stream(), logits.data_ptr<float>(),
+  auto stream = current_stream();
+  hipLaunchKernelGGL(categorical_policy_loss_bwd, dim3(n_blocks), dim3(256), 0,
+                     stream, logits.data_ptr<float>(),
                      actions.data_ptr<float>(), olp, adv.data_ptr<float>(),
-                     dlogits.data_ptr<float>(), scalars.data_ptr<float>(), B, N,
+                     dlogits.data_ptr<float>(), partials.data_ptr<float>(), B, N,
                      (float)clip, (int)mode);
+  HIP_OK(hipGetLastError());
+  hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(1), 0, stream,
+                     partials.data_ptr<float>(), nullptr,
+                     scalars.data_ptr<float>(), n_blocks, 0);
   HIP_OK(hipGetLastError());
   return {dlogits, scalars};
 }
@@ -337,7 +349,8 @@ std::vector<torch::Tensor> value_mse_loss(torch::Tensor v, torch::Tensor ret) {
   check_f32_gpu(v, "v");
   const int B = (int)v.numel();
   auto dv = torch::empty_like(v);
-  auto scalars = torch::empty({1}, v.options());
+  // scalars[0] = loss; scalars[1] accumulates across calls (graph loops)
+  auto scalars = torch::zeros({2}, v.options());
   hipLaunchKernelGGL(value_mse_bwd_kernel, dim3(1), dim3(1024), 0,
                      current_stream(), v.data_ptr<float>(), ret.data_ptr<float>(),
                      dv.data_ptr<float>(), scalars.data_ptr<float>(), B);

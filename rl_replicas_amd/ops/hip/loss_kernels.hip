@@ -60,34 +60,48 @@ DEV_INLINE float dlogp_coeff(int mode, float logp, float old_logp, float adv,
 // ---------------------------------------------------------------------------
 // Gaussian policy (state-independent log_std — gaussian_policy.py:18-35):
 // logp(a|s) = sum_d [ -0.5 z_d^2 - log_std_d ] - D/2 log(2pi),  z=(a-mean)/sigma
-// outputs: dmean[B,D], dlog_std[D], scalars[0]=loss
+//
+// Multi-block phase 1 (templated compile-time D -> register-resident
+// dlog_std accumulators) writing per-block partials; a tiny finalize
+// kernel sums the <=LOSS_BLOCKS partials in fixed order (deterministic).
+// outputs: dmean[B,D]; partials[blk][D+1] = (dlog_std_d..., loss)
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(LOSS_THREADS) void gaussian_policy_loss_bwd(
+#define LOSS_BLOCKS 32
+#define LOSS_P_THREADS 256
+
+template <int DT>
+__global__ __launch_bounds__(LOSS_P_THREADS) void gaussian_policy_loss_bwd_t(
     const float* __restrict__ mean, const float* __restrict__ actions,
     const float* __restrict__ old_logp, const float* __restrict__ adv,
     const float* __restrict__ log_std, float* __restrict__ dmean,
-    float* __restrict__ dlog_std, float* __restrict__ coef_ws,
-    float* __restrict__ scalars, int B, int D, float clip, int mode) {
-  __shared__ float red[LOSS_THREADS / WAVE];
-  __shared__ float s_sigma[64];    // sigma, 1/sigma^2 cached (D <= 32)
+    float* __restrict__ partials, int B, float clip, int mode) {
+  __shared__ float red[LOSS_P_THREADS / WAVE];
   const int tid = threadIdx.x;
-  if (tid < D) {
-    float s = __expf(log_std[tid]);
-    s_sigma[tid] = s;
-    s_sigma[32 + tid] = 1.f / (s * s);
-  }
-  __syncthreads();
+  const int gid = blockIdx.x * LOSS_P_THREADS + tid;
 
-  float base = 0.f;
-  for (int d = 0; d < D; ++d) base += log_std[d];
-  base += 0.5f * (float)D * LOG_2PI;
+  float sigma[DT], inv_s2[DT];
+  float base = 0.5f * (float)DT * LOG_2PI;
+  #pragma unroll
+  for (int d = 0; d < DT; ++d) {
+    const float ls = log_std[d];
+    sigma[d] = __expf(ls);
+    inv_s2[d] = 1.f / (sigma[d] * sigma[d]);
+    base += ls;
+  }
 
   const float inv_b = 1.f / (float)B;
   float loss_acc = 0.f;
-  for (int r = tid; r < B; r += LOSS_THREADS) {
+  float dls[DT];
+  #pragma unroll
+  for (int d = 0; d < DT; ++d) dls[d] = 0.f;
+
+  for (int r = gid; r < B; r += gridDim.x * LOSS_P_THREADS) {
+    float diff[DT];
     float q = 0.f;
-    for (int d = 0; d < D; ++d) {
-      const float z = (actions[(long)r * D + d] - mean[(long)r * D + d]) / s_sigma[d];
+    #pragma unroll
+    for (int d = 0; d < DT; ++d) {
+      diff[d] = actions[(long)r * DT + d] - mean[(long)r * DT + d];
+      const float z = diff[d] / sigma[d];
       q += z * z;
     }
     const float logp = -0.5f * q - base;
@@ -95,28 +109,51 @@ __global__ __launch_bounds__(LOSS_THREADS) void gaussian_policy_loss_bwd(
     const float c = dlogp_coeff(mode, logp, old_logp ? old_logp[r] : 0.f,
                                 adv[r], clip, &loss_r) * inv_b;
     loss_acc += loss_r * inv_b;
-    coef_ws[r] = c;
-    for (int d = 0; d < D; ++d) {
-      const float diff = actions[(long)r * D + d] - mean[(long)r * D + d];
-      // dlogp/dmean_d = (a_d - mean_d)/sigma_d^2
-      dmean[(long)r * D + d] = c * diff * s_sigma[32 + d];
+    #pragma unroll
+    for (int d = 0; d < DT; ++d) {
+      dmean[(long)r * DT + d] = c * diff[d] * inv_s2[d];
+      const float z2 = diff[d] * diff[d] * inv_s2[d];
+      dls[d] += c * (z2 - 1.f);
     }
   }
-  const float loss = block_sum(loss_acc, red);
-  if (tid == 0) scalars[0] = loss;
 
-  // dlog_std_d = sum_r c_r * (z_d^2 - 1); re-read per d (D is tiny)
-  __threadfence_block();
-  __syncthreads();
-  for (int d = 0; d < D; ++d) {
-    float acc = 0.f;
-    for (int r = tid; r < B; r += LOSS_THREADS) {
-      const float z = (actions[(long)r * D + d] - mean[(long)r * D + d]) / s_sigma[d];
-      acc += coef_ws[r] * (z * z - 1.f);
+  // per-block fixed-order reductions -> partials row
+  #pragma unroll
+  for (int d = 0; d < DT; ++d) {
+    float v = wave_reduce_sum(dls[d]);
+    if ((tid & 63) == 0) red[tid / WAVE] = v;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < LOSS_P_THREADS / WAVE; ++w) t += red[w];
+      partials[(long)blockIdx.x * (DT + 1) + d] = t;
     }
-    const float total = block_sum(acc, red);
-    if (tid == 0) dlog_std[d] = total;
+    __syncthreads();
   }
+  {
+    float v = wave_reduce_sum(loss_acc);
+    if ((tid & 63) == 0) red[tid / WAVE] = v;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < LOSS_P_THREADS / WAVE; ++w) t += red[w];
+      partials[(long)blockIdx.x * (DT + 1) + DT] = t;
+    }
+  }
+}
+
+// sum the per-block partial rows in fixed order:
+// dlog_std[d] = sum_blk partials[blk][d]; scalars[0] = sum_blk partials[blk][D]
+__global__ void loss_partials_finalize(const float* __restrict__ partials,
+                                       float* __restrict__ dlog_std,
+                                       float* __restrict__ scalars, int n_blocks,
+                                       int D) {
+  const int d = threadIdx.x;
+  if (d > D) return;
+  float s = 0.f;
+  for (int p = 0; p < n_blocks; ++p) s += partials[(long)p * (D + 1) + d];
+  if (d < D) dlog_std[d] = s;
+  else scalars[0] = s;
 }
 
 // logp only (old-policy snapshot at epoch start)
@@ -181,16 +218,17 @@ DEV_INLINE float row_lse(const float* logits, int n) {
   return m + __logf(s);
 }
 
-__global__ __launch_bounds__(LOSS_THREADS) void categorical_policy_loss_bwd(
+__global__ __launch_bounds__(LOSS_P_THREADS) void categorical_policy_loss_bwd(
     const float* __restrict__ logits, const float* __restrict__ actions,
     const float* __restrict__ old_logp, const float* __restrict__ adv,
-    float* __restrict__ dlogits, float* __restrict__ scalars, int B, int N,
+    float* __restrict__ dlogits, float* __restrict__ partials, int B, int N,
     float clip, int mode) {
-  __shared__ float red[LOSS_THREADS / WAVE];
+  __shared__ float red[LOSS_P_THREADS / WAVE];
   const int tid = threadIdx.x;
   const float inv_b = 1.f / (float)B;
   float loss_acc = 0.f;
-  for (int r = tid; r < B; r += LOSS_THREADS) {
+  for (int r = blockIdx.x * LOSS_P_THREADS + tid; r < B;
+       r += gridDim.x * LOSS_P_THREADS) {
     const float* lg = logits + (long)r * N;
     const int a = (int)actions[r];
     const float lse = row_lse(lg, N);
@@ -204,8 +242,14 @@ __global__ __launch_bounds__(LOSS_THREADS) void categorical_policy_loss_bwd(
       dlogits[(long)r * N + j] = c * (((j == a) ? 1.f : 0.f) - p);
     }
   }
-  const float loss = block_sum(loss_acc, red);
-  if (tid == 0) scalars[0] = loss;
+  float v = wave_reduce_sum(loss_acc);
+  if ((tid & 63) == 0) red[tid / WAVE] = v;
+  __syncthreads();
+  if (tid == 0) {
+    float t = 0.f;
+    for (int w = 0; w < LOSS_P_THREADS / WAVE; ++w) t += red[w];
+    partials[blockIdx.x] = t;  // rows of size 1 (D=0) for the finalizer
+  }
 }
 
 __global__ __launch_bounds__(LOSS_THREADS) void categorical_logp_kernel(
@@ -232,6 +276,27 @@ __global__ __launch_bounds__(LOSS_THREADS) void categorical_kl_kernel(
   if (tid == 0) out[0] = total / (float)B;
 }
 
+// host-side dispatch over the compile-time-D instantiations
+void launch_gaussian_loss(const float* mean, const float* actions,
+                          const float* old_logp, const float* adv,
+                          const float* log_std, float* dmean, float* partials,
+                          int B, int D, float clip, int mode, int n_blocks,
+                          hipStream_t stream) {
+  dim3 g(n_blocks), b(LOSS_P_THREADS);
+#define CASE_D(DT)                                                           \
+  case DT:                                                                   \
+    hipLaunchKernelGGL((gaussian_policy_loss_bwd_t<DT>), g, b, 0, stream,    \
+                       mean, actions, old_logp, adv, log_std, dmean,         \
+                       partials, B, clip, mode);                             \
+    break;
+  switch (D) {
+    CASE_D(1) CASE_D(2) CASE_D(3) CASE_D(4)
+    CASE_D(5) CASE_D(6) CASE_D(7) CASE_D(8)
+    default: break;  // guarded by the binding (D <= 8)
+  }
+#undef CASE_D
+}
+
 // ---------------------------------------------------------------------------
 // value MSE: loss = mean((v - ret)^2), dv = 2(v - ret)/B  (ppo.py:283-287)
 // ---------------------------------------------------------------------------
@@ -248,5 +313,8 @@ __global__ __launch_bounds__(LOSS_THREADS) void value_mse_bwd_kernel(
     dv[r] = 2.f * diff * inv_b;
   }
   const float loss = block_sum(loss_acc, red);
-  if (tid == 0) scalars[0] = loss;
+  if (tid == 0) {
+    scalars[0] = loss;
+    scalars[1] += loss;  // device-side accumulator for graph-replayed loops
+  }
 }

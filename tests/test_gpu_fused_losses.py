@@ -207,3 +207,75 @@ class TestFusedUpdateLoopEquivalence:
         assert abs(r_fused["policy/kl_divergence"] - r_eager["policy/kl_divergence"]) < 1e-4
         for p1, p2 in zip(m1.policy.parameters(), m2.policy.parameters()):
             torch.testing.assert_close(p1, p2, rtol=1e-3, atol=1e-5)
+
+
+class TestGraphedLoops:
+    def _make_ppo(self, seed=0):
+        import torch.nn as nn
+
+        from rl_replicas_amd import envs, ops
+        from rl_replicas_amd.algorithms import PPO
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import GaussianPolicy
+        from rl_replicas_amd.samplers import VectorSampler
+        from rl_replicas_amd.value_function import ValueFunction
+
+        torch.manual_seed(seed)
+        pnet = MLP([17, 64, 32, 6]).to("cuda")
+        log_std = nn.Parameter(-0.5 * torch.ones(6, device="cuda"))
+        policy = GaussianPolicy(
+            pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+        )
+        vnet = MLP([17, 64, 32, 1]).to("cuda")
+        vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+        venv = envs.VectorEnv("HalfCheetah-v4", num_envs=10)
+        return PPO(policy, vf, venv, VectorSampler(venv, seed=0),
+                   num_policy_gradients=5, num_value_gradients=5)
+
+    def test_graphed_equals_eager_fused(self, ext, monkeypatch):
+        """hipGraph-captured policy+value loops produce the same params
+        as the eager fused path."""
+        from rl_replicas_amd.ops import fused_onpolicy as fop
+        from rl_replicas_amd.ops.fused_adam import FusedAdam
+
+        torch.manual_seed(42)
+        obs = torch.randn(1000, 17, device="cuda")
+        actions = torch.randn(1000, 6, device="cuda")
+        adv = torch.randn(1000, device="cuda")
+        returns = torch.randn(1000, device="cuda")
+
+        m1 = self._make_ppo(0)
+        m2 = self._make_ppo(0)
+        assert isinstance(m1.policy.optimizer, FusedAdam)
+
+        monkeypatch.setenv("RL_REPLICAS_AMD_DISABLE_GRAPHS", "1")
+        r_eager = fop.ppo_update(m2, obs, actions, adv)
+        l_eager = fop.value_update(m2, obs, returns, 5)
+        monkeypatch.delenv("RL_REPLICAS_AMD_DISABLE_GRAPHS")
+
+        r_graph = fop.ppo_update(m1, obs, actions, adv)
+        l_graph = fop.value_update(m1, obs, returns, 5)
+
+        assert abs(r_graph["policy/loss"] - r_eager["policy/loss"]) < 1e-5
+        assert abs(r_graph["policy/kl_divergence"] - r_eager["policy/kl_divergence"]) < 1e-5
+        assert abs(l_graph - l_eager) < 1e-4
+        for p1, p2 in zip(m1.policy.parameters(), m2.policy.parameters()):
+            torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-7)
+        for p1, p2 in zip(m1.value_function.parameters(), m2.value_function.parameters()):
+            torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-7)
+
+    def test_graph_replay_across_epochs(self, ext):
+        """Second epoch reuses the cached graph with new data; params
+        keep evolving and stay finite."""
+        from rl_replicas_amd.ops import fused_onpolicy as fop
+
+        m = self._make_ppo(1)
+        for _ in range(3):
+            obs = torch.randn(1000, 17, device="cuda")
+            actions = torch.randn(1000, 6, device="cuda")
+            adv = torch.randn(1000, device="cuda")
+            fop.ppo_update(m, obs, actions, adv)
+            fop.value_update(m, obs, torch.randn(1000, device="cuda"), 5)
+        assert getattr(m, "_ppo_policy_graph", None) is not None
+        for p in m.policy.parameters():
+            assert torch.isfinite(p).all()
