@@ -262,7 +262,7 @@ class _GraphedValueLoop:
             for _ in range(num_iters):
                 out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
                 dv, scalars = ext.value_mse_loss(out.view(-1), self.returns)
-                losses.append(scalars[:1])
+                losses.append(scalars)
                 _backward_and_step(
                     vf, mlp, self.obs, dv.view(out.shape), hidden, out, weights,
                     biases, acts, [], lambda _m: None,
@@ -421,7 +421,7 @@ def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
     for _ in range(num_iters):
         out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
         dv, scalars = ext.value_mse_loss(out.view(-1), returns)
-        losses.append(scalars[:1])
+        losses.append(scalars)
         _backward_and_step(
             vf, mlp, obs, dv.view(out.shape), hidden, out, weights, biases, acts,
             [], algo._all_reduce_gradients,

@@ -15,11 +15,15 @@
 
 // kernel declarations (defined in the .hip translation units)
 void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
-                    int rows, int maxw, int n_blocks, hipStream_t stream);
+                    int rows, int maxw, int n_blocks, int wstage_mode,
+                    size_t lds_bytes, hipStream_t stream);
 void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
-                          const float* W, float* dx, float* ws, int batch,
-                          int out_d, int in_d, int act, int rows, int maxw,
-                          int n_blocks, hipStream_t stream);
+                          const float* W, float* dx, float* ws, long ws_stride,
+                          int batch, int out_d, int in_d, int act, int rows,
+                          int maxw, int n_blocks, int wstage_mode,
+                          size_t lds_bytes, hipStream_t stream);
+__global__ void mlp_grad_reduce_stage_f32(const float* ws, float* out,
+                                          int n_blocks, int chunk, long grand);
 __global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a);
 void launch_gaussian_loss(const float* mean, const float* actions,
                           const float* old_logp, const float* adv,
@@ -88,12 +92,42 @@ void check_f32_gpu(const torch::Tensor& t, const char* name) {
 // (4000-row batch -> 125 workgroups); 64 otherwise.
 void pick_tile(int batch, int max_width, int* rows, int* maxw) {
   *maxw = max_width <= 64 ? 64 : 256;
-  *rows = batch >= 2048 ? 32 : 64;
+  *rows = (batch >= 2048 && *maxw == 64) ? 32 : (*maxw == 256 ? 32 : 64);
   static int env_rows = []() {
     const char* e = getenv("RL_REPLICAS_AMD_MLP_ROWS");
     return e ? atoi(e) : 0;
   }();
   if (env_rows == 32 || env_rows == 64) *rows = env_rows;
+  if (*maxw == 256) *rows = 32;  // only <32,256> is instantiated
+}
+
+#define KCHUNK 64
+
+// weight-staging mode + dynamic-LDS size for the fused MLP kernels:
+// mode 0 stages the whole net per block (narrow nets), mode 1 stages
+// per-wave W sub-tiles.  act_elems = the two activation ping-pong
+// buffers (fwd) or dz+xt (bwd) — same footprint.
+void pick_wstage(const std::vector<torch::Tensor>& weights, int rows, int maxw,
+                 bool backward_single_layer, int layer, int* mode,
+                 size_t* lds_bytes) {
+  const int ldsw = maxw + 4;
+  size_t act_elems = (size_t)2 * rows * ldsw;
+  size_t whole = 0;
+  if (backward_single_layer) {
+    whole = (size_t)weights[layer].size(0) * (weights[layer].size(1) + 1);
+  } else {
+    for (auto& w : weights) whole += (size_t)w.size(0) * (w.size(1) + 1);
+  }
+  size_t mode0_bytes = (act_elems + whole) * 4;
+  if (mode0_bytes <= 100 * 1024) {
+    *mode = 0;
+    *lds_bytes = mode0_bytes;
+  } else {
+    *mode = 1;
+    size_t wv = backward_single_layer ? (size_t)4 * KCHUNK * 18
+                                      : (size_t)4 * 16 * (KCHUNK + 2);
+    *lds_bytes = (act_elems + wv) * 4;
+  }
 }
 
 }  // namespace
@@ -147,9 +181,12 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
   int rows, maxw;
   pick_tile(args.batch, max_width, &rows, &maxw);
   const int n_blocks = (args.batch + rows - 1) / rows;
+  int wmode;
+  size_t lds_bytes;
+  pick_wstage(weights, rows, maxw, false, 0, &wmode, &lds_bytes);
   if (n_blocks > 0) {
     launch_mlp_fwd(args, x.data_ptr<float>(), save_hidden ? 1 : 0, rows, maxw,
-                   n_blocks, current_stream());
+                   n_blocks, wmode, lds_bytes, current_stream());
     HIP_OK(hipGetLastError());
   }
   return outs;
@@ -174,17 +211,17 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
   auto opts = x.options();
   auto stream = current_stream();
 
-  // one workspace for every layer's per-row-block partials
-  std::vector<int64_t> totals(L), ws_off(L);
-  int64_t ws_elems = 0;
+  // one workspace: [block][concatenated layer elems] (layer-blind stage-1)
+  std::vector<int64_t> totals(L), layer_off(L);
+  int64_t grand = 0;
   for (int l = 0; l < L; ++l) {
     const int out_d = (int)weights[l].size(0);
     const int in_d = (int)weights[l].size(1);
     totals[l] = (int64_t)out_d * in_d + out_d;
-    ws_off[l] = ws_elems;
-    ws_elems += (int64_t)n_blocks * totals[l];
+    layer_off[l] = grand;
+    grand += totals[l];
   }
-  torch::Tensor ws = torch::empty({ws_elems}, opts);
+  torch::Tensor ws = torch::empty({(int64_t)n_blocks, grand}, opts);
   float* ws_ptr = ws.data_ptr<float>();
 
   std::vector<torch::Tensor> dws(L), dbs(L);
@@ -198,27 +235,44 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dws[l] = torch::empty({out_d, in_d}, opts);
     dbs[l] = torch::empty({out_d}, opts);
     dx = torch::empty({batch, in_d}, opts);
+    int wmode;
+    size_t lds_bytes;
+    pick_wstage(weights, rows, maxw, true, l, &wmode, &lds_bytes);
     // merged dgrad + wgrad/bias partials in one kernel
     launch_mlp_bwd_layer(dy.data_ptr<float>(), y.data_ptr<float>(),
                          xin.data_ptr<float>(), weights[l].data_ptr<float>(),
-                         dx.data_ptr<float>(), ws_ptr + ws_off[l], batch, out_d,
-                         in_d, (int)acts[l], rows, maxw, n_blocks, stream);
+                         dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
+                         batch, out_d, in_d, (int)acts[l], rows, maxw, n_blocks,
+                         wmode, lds_bytes, stream);
     HIP_OK(hipGetLastError());
     dy = dx;
   }
 
-  // single deterministic reduction launch over every layer's partials
+  // two-stage deterministic reduction over the partial rows
+  const float* red_src = ws_ptr;
+  int red_n = n_blocks;
+  torch::Tensor ws2;
+  if (n_blocks > 16) {
+    const int chunks = 8;
+    const int chunk = (n_blocks + chunks - 1) / chunks;
+    ws2 = torch::empty({(int64_t)chunks, grand}, opts);
+    dim3 g((unsigned)std::min<int64_t>(128, (grand + 255) / 256), chunks);
+    hipLaunchKernelGGL(mlp_grad_reduce_stage_f32, g, dim3(256), 0, stream,
+                       ws_ptr, ws2.data_ptr<float>(), n_blocks, chunk, grand);
+    HIP_OK(hipGetLastError());
+    red_src = ws2.data_ptr<float>();
+    red_n = chunks;
+  }
   ReduceAllArgs ra{};
+  ra.ws = red_src;
+  ra.stride = grand;
   ra.n_layers = L;
-  ra.n_blocks = n_blocks;
-  int64_t grand = 0;
+  ra.n_blocks = red_n;
   for (int l = 0; l < L; ++l) {
-    ra.ws[l] = ws_ptr + ws_off[l];
     ra.dw[l] = dws[l].data_ptr<float>();
     ra.db[l] = dbs[l].data_ptr<float>();
     ra.total[l] = (int)totals[l];
     ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
-    grand += totals[l];
   }
   int rb = (int)std::min<int64_t>(512, (grand + 255) / 256);
   hipLaunchKernelGGL(mlp_grad_reduce_all_f32, dim3(rb), dim3(256), 0, stream, ra);
@@ -349,8 +403,7 @@ std::vector<torch::Tensor> value_mse_loss(torch::Tensor v, torch::Tensor ret) {
   check_f32_gpu(v, "v");
   const int B = (int)v.numel();
   auto dv = torch::empty_like(v);
-  // scalars[0] = loss; scalars[1] accumulates across calls (graph loops)
-  auto scalars = torch::zeros({2}, v.options());
+  auto scalars = torch::empty({1}, v.options());
   hipLaunchKernelGGL(value_mse_bwd_kernel, dim3(1), dim3(1024), 0,
                      current_stream(), v.data_ptr<float>(), ret.data_ptr<float>(),
                      dv.data_ptr<float>(), scalars.data_ptr<float>(), B);

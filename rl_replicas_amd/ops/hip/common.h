@@ -56,9 +56,12 @@ struct MLPArgs {
   int batch;
 };
 
-// all-layer gradient-reduction argument block (mlp_kernels.hip)
+// all-layer gradient-reduction argument block (mlp_kernels.hip).
+// workspace layout: ws[partial][flat-elem] with the layers' (od*id+od)
+// segments concatenated in layer order inside each partial row.
 struct ReduceAllArgs {
-  const float* ws[MLP_MAX_LAYERS];
+  const float* ws;            // [n_blocks][stride]
+  long stride;                // flat elems per partial row (grand total)
   float* dw[MLP_MAX_LAYERS];
   float* db[MLP_MAX_LAYERS];
   int total[MLP_MAX_LAYERS];  // od*id + od per layer

@@ -313,8 +313,5 @@ __global__ __launch_bounds__(LOSS_THREADS) void value_mse_bwd_kernel(
     dv[r] = 2.f * diff * inv_b;
   }
   const float loss = block_sum(loss_acc, red);
-  if (tid == 0) {
-    scalars[0] = loss;
-    scalars[1] += loss;  // device-side accumulator for graph-replayed loops
-  }
+  if (tid == 0) scalars[0] = loss;
 }

@@ -5,32 +5,35 @@
 // tiny-MLP / launch-latency-bound regime this library lives in
 // (SURVEY.md §7 "Tiny-tensor regime"): the whole multi-layer forward is
 // ONE kernel launch; each workgroup owns a ROWS-row tile of the batch,
-// ping-pongs layer activations between two padded LDS buffers, and runs
+// ping-pongs layer activations between two padded LDS regions, and runs
 // every Linear layer as MFMA tiles (v_mfma_f32_16x16x4_f32: exact fp32,
 // guide §3) with bias+activation fused into the epilogue.
 //
-// Kernels are templated on <ROWS, MAXW>:
-//   MAXW = widest supported layer (64 for the on-policy nets, 256 for
-//          the off-policy nets) -> LDS footprint 2*ROWS*(MAXW+4)*4 B,
-//          so narrow nets run at multi-block-per-CU occupancy;
-//   ROWS = batch rows per workgroup (32 fills the 256-CU chip at the
-//          reference's 4000-row batches: 125 blocks; 64 for small
-//          batches).  4 waves split (row-tile x output-tile) work.
+// Weight access is the latency hazard at these sizes (a scattered
+// per-lane W read per MFMA serializes on L2 latency when the grid is
+// small), so weights are staged through LDS:
+//   mode 0 (narrow nets): the ENTIRE net's weights are staged once per
+//     block into a padded LDS image ([out][(in+1)] per layer -> odd row
+//     stride, bank-conflict-free for both W and W^T reads);
+//   mode 1 (wide nets): each wave stages per-(output-tile, K-chunk)
+//     sub-tiles of W into its private LDS slice, synchronized with
+//     wave-local lgkmcnt waits (no cross-wave barriers).
+// The host picks the mode from the LDS budget (dynamic shared memory).
 //
 // Backward = ONE merged kernel per layer (dgrad + wgrad/bias partials
-// sharing the staged dZ tile) + ONE deterministic all-layer partial
+// sharing the staged dZ tile) + a TWO-STAGE deterministic partial
 // reduction (split-K via workspace instead of atomics so every run is
-// bitwise reproducible — the framework's determinism contract,
-// SURVEY.md §4).
+// bitwise reproducible — SURVEY.md §4).  Workspace layout is
+// [block][all-layer elems] so the first reduction stage is layer-blind.
 #include "common.h"
 
 // ---------------------------------------------------------------------------
 // helpers
 // ---------------------------------------------------------------------------
-template <int ROWS, int LDSW>
+template <int LDSW>
 DEV_INLINE void load_tile(const float* __restrict__ src, float* dst_lds,
-                          int row0, int batch, int width, int tid) {
-  for (int idx = tid; idx < ROWS * width; idx += 256) {
+                          int row0, int batch, int width, int tid, int rows) {
+  for (int idx = tid; idx < rows * width; idx += 256) {
     int r = idx / width, c = idx % width;
     float v = 0.f;
     int row = row0 + r;
@@ -39,39 +42,69 @@ DEV_INLINE void load_tile(const float* __restrict__ src, float* dst_lds,
   }
 }
 
-template <int ROWS, int LDSW>
+template <int LDSW>
 DEV_INLINE void store_tile(float* __restrict__ dst, const float* src_lds,
-                           int row0, int batch, int width, int tid) {
-  for (int idx = tid; idx < ROWS * width; idx += 256) {
+                           int row0, int batch, int width, int tid, int rows) {
+  for (int idx = tid; idx < rows * width; idx += 256) {
     int r = idx / width, c = idx % width;
     int row = row0 + r;
     if (row < batch) dst[(long)row * width + c] = src_lds[r * LDSW + c];
   }
 }
 
+// stage W[rows0..rows0+nrows)[0..ncols) -> lds image with row stride
+// (ncols+1) — whole-block cooperative, caller barriers
+DEV_INLINE void stage_weights_block(const float* __restrict__ W, float* wlds,
+                                    int nrows, int ncols, int tid) {
+  for (int idx = tid; idx < nrows * ncols; idx += 256) {
+    int r = idx / ncols, c = idx % ncols;
+    wlds[r * (ncols + 1) + c] = W[(long)r * ncols + c];
+  }
+}
+
+// wave-local LDS write->read ordering (all writers are this wave's lanes)
+DEV_INLINE void wave_lds_fence() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+}
+
+#define KCHUNK 64  // K-chunk depth for mode-1 weight staging
+
 // ---------------------------------------------------------------------------
 // fused forward
 // ---------------------------------------------------------------------------
 template <int ROWS, int MAXW>
 __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
-    MLPArgs args, const float* __restrict__ x, int save_hidden) {
+    MLPArgs args, const float* __restrict__ x, int save_hidden, int wstage_mode) {
   constexpr int LDSW = MAXW + 4;
   constexpr int RT = ROWS / 16;        // row tiles per block
   constexpr int JT_STRIDE = 4 / RT;    // waves sharing one row tile
-  __shared__ float buf[2][ROWS * LDSW];
+  extern __shared__ float smem[];
+  float* const buf0 = smem;
+  float* const buf1 = smem + ROWS * LDSW;
+  float* wlds = smem + 2 * ROWS * LDSW;  // mode 0: whole net; mode 1: per-wave
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int row0 = blockIdx.x * ROWS;
   const int wr0 = (wave % RT) * 16;
   const int jt0 = (wave / RT) * 16;
+  // mode-1 per-wave W slice: [16][KCHUNK+2]
+  float* wv = wlds + wave * 16 * (KCHUNK + 2);
 
-  load_tile<ROWS, LDSW>(x, buf[0], row0, args.batch, args.dims[0], tid);
+  load_tile<LDSW>(x, buf0, row0, args.batch, args.dims[0], tid, ROWS);
+  if (wstage_mode == 0) {
+    int off = 0;
+    for (int l = 0; l < args.n_layers; ++l) {
+      stage_weights_block(args.w[l], wlds + off, args.dims[l + 1], args.dims[l], tid);
+      off += args.dims[l + 1] * (args.dims[l] + 1);
+    }
+  }
   __syncthreads();
 
   const int i = lane & 15;
   const int k = lane >> 4;
   int cur = 0;
+  int woff = 0;
   for (int l = 0; l < args.n_layers; ++l) {
     const int in_d = args.dims[l];
     const int out_d = args.dims[l + 1];
@@ -79,23 +112,47 @@ __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
     const float* B = args.b[l];
     const int act = args.acts[l];
     const int nxt = cur ^ 1;
+    const float* buf_in = (cur == 0) ? buf0 : buf1;
+    float* buf_out = (cur == 0) ? buf1 : buf0;
+    const int wrow = in_d + 1;  // mode-0 padded W row stride
 
     for (int jt = jt0; jt < out_d; jt += 16 * JT_STRIDE) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
       const int j = jt + i;
       const bool jok = j < out_d;
-      for (int k0 = 0; k0 < in_d; k0 += 4) {
-        const int kk = k0 + k;
-        float a = (kk < in_d) ? buf[cur][(wr0 + i) * LDSW + kk] : 0.f;
-        float bv = (jok && kk < in_d) ? W[(long)j * in_d + kk] : 0.f;
-        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+      if (wstage_mode == 0) {
+        const float* wl = wlds + woff;
+        for (int k0 = 0; k0 < in_d; k0 += 4) {
+          const int kk = k0 + k;
+          float a = (kk < in_d) ? buf_in[(wr0 + i) * LDSW + kk] : 0.f;
+          float bv = (jok && kk < in_d) ? wl[j * wrow + kk] : 0.f;
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
+      } else {
+        // per-wave staged K-chunks of W[jt..jt+16)[c0..c0+KCHUNK)
+        for (int c0 = 0; c0 < in_d; c0 += KCHUNK) {
+          const int clen = min(KCHUNK, in_d - c0);
+          for (int idx = lane; idx < 16 * clen; idx += 64) {
+            int r = idx / clen, c = idx % clen;
+            float v = (jt + r < out_d) ? W[(long)(jt + r) * in_d + c0 + c] : 0.f;
+            wv[r * (KCHUNK + 2) + c] = v;
+          }
+          wave_lds_fence();
+          for (int k0 = 0; k0 < clen; k0 += 4) {
+            const int kk = k0 + k;
+            float a = (kk < clen) ? buf_in[(wr0 + i) * LDSW + c0 + kk] : 0.f;
+            float bv = (kk < clen) ? wv[i * (KCHUNK + 2) + kk] : 0.f;
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+          }
+          wave_lds_fence();  // reads done before next chunk overwrites
+        }
       }
       if (jok) {
         const float bias = B[j];
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = wr0 + (lane >> 4) * 4 + r;
-          buf[nxt][row * LDSW + j] = act_apply(act, acc[r] + bias);
+          buf_out[row * LDSW + j] = act_apply(act, acc[r] + bias);
         }
       }
     }
@@ -103,10 +160,11 @@ __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
 
     const bool is_last = (l == args.n_layers - 1);
     if (is_last || save_hidden) {
-      store_tile<ROWS, LDSW>(args.h[l], buf[nxt], row0, args.batch, out_d, tid);
+      store_tile<LDSW>(args.h[l], buf_out, row0, args.batch, out_d, tid, ROWS);
     }
     cur = nxt;
-    // next layer writes buf[cur^1] (fully consumed) and reads buf[cur]
+    woff += out_d * wrow;
+    // next layer writes bufs[cur^1] (fully consumed) and reads bufs[cur]
     // (fully written before the barrier above) -> one barrier per layer
   }
 }
@@ -116,25 +174,29 @@ __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
 //   dZ = dY * act'(y)           (staged once in LDS)
 //   dX = dZ @ W                 (row tiles, written to global)
 //   dW_p = dZ^T @ X, db_p       (per-row-block partials to workspace)
+// workspace layout: ws[block][layer-elems at layer_off]
 // ---------------------------------------------------------------------------
 template <int ROWS, int MAXW>
 __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
     const float* __restrict__ dy, const float* __restrict__ y,
     const float* __restrict__ xin, const float* __restrict__ W,
-    float* __restrict__ dx, float* __restrict__ workspace, int batch,
-    int out_d, int in_d, int act) {
+    float* __restrict__ dx, float* __restrict__ workspace, long ws_stride,
+    int batch, int out_d, int in_d, int act, int wstage_mode) {
   constexpr int LDSW = MAXW + 4;
   constexpr int RT = ROWS / 16;
   constexpr int JT_STRIDE = 4 / RT;
-  __shared__ float dz[ROWS * LDSW];
-  __shared__ float xt[ROWS * LDSW];
+  extern __shared__ float smem[];
+  float* dz = smem;
+  float* xt = smem + ROWS * LDSW;
+  float* wlds = smem + 2 * ROWS * LDSW;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int row0 = blockIdx.x * ROWS;
   const int wr0 = (wave % RT) * 16;
   const int jt0 = (wave / RT) * 16;
-  float* wsp = workspace + (long)blockIdx.x * (out_d * in_d + out_d);
+  float* wv = wlds + wave * KCHUNK * 18;  // mode-1 slice: [KCHUNK][16+2]
+  float* wsp = workspace + (long)blockIdx.x * ws_stride;
 
   for (int idx = tid; idx < ROWS * out_d; idx += 256) {
     int r = idx / out_d, c = idx % out_d;
@@ -146,22 +208,46 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
     }
     dz[r * LDSW + c] = v;
   }
-  load_tile<ROWS, LDSW>(xin, xt, row0, batch, in_d, tid);
+  load_tile<LDSW>(xin, xt, row0, batch, in_d, tid, ROWS);
+  if (wstage_mode == 0) {
+    stage_weights_block(W, wlds, out_d, in_d, tid);
+  }
   __syncthreads();
 
   const int i = lane & 15;
   const int k = lane >> 4;
+  const int wrow = in_d + 1;
 
   // ---- dgrad: dX[b][j] = sum_k dZ[b][k] W[k][j] ----
   for (int jt = jt0; jt < in_d; jt += 16 * JT_STRIDE) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     const int j = jt + i;
     const bool jok = j < in_d;
-    for (int k0 = 0; k0 < out_d; k0 += 4) {
-      const int kk = k0 + k;
-      float a = (kk < out_d) ? dz[(wr0 + i) * LDSW + kk] : 0.f;
-      float bv = (jok && kk < out_d) ? W[(long)kk * in_d + j] : 0.f;
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    if (wstage_mode == 0) {
+      for (int k0 = 0; k0 < out_d; k0 += 4) {
+        const int kk = k0 + k;
+        float a = (kk < out_d) ? dz[(wr0 + i) * LDSW + kk] : 0.f;
+        float bv = (jok && kk < out_d) ? wlds[kk * wrow + j] : 0.f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+      }
+    } else {
+      // per-wave staged chunks of W[c0..c0+KCHUNK)[jt..jt+16)
+      for (int c0 = 0; c0 < out_d; c0 += KCHUNK) {
+        const int clen = min(KCHUNK, out_d - c0);
+        for (int idx = lane; idx < clen * 16; idx += 64) {
+          int r = idx / 16, c = idx % 16;
+          float v = (jt + c < in_d) ? W[(long)(c0 + r) * in_d + jt + c] : 0.f;
+          wv[r * 18 + c] = v;
+        }
+        wave_lds_fence();
+        for (int k0 = 0; k0 < clen; k0 += 4) {
+          const int kk = k0 + k;
+          float a = (kk < clen) ? dz[(wr0 + i) * LDSW + c0 + kk] : 0.f;
+          float bv = (kk < clen) ? wv[kk * 18 + i] : 0.f;
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
+        wave_lds_fence();
+      }
     }
     if (jok) {
       #pragma unroll
@@ -206,38 +292,58 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
 // host-side dispatch over the (ROWS, MAXW) instantiations — called from
 // bindings.hip so template symbols stay in this translation unit
 void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
-                    int rows, int maxw, int n_blocks, hipStream_t stream) {
+                    int rows, int maxw, int n_blocks, int wstage_mode,
+                    size_t lds_bytes, hipStream_t stream) {
   dim3 g(n_blocks), b(256);
   if (rows == 32 && maxw == 64)
-    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<32, 64>), g, b, 0, stream, args, x, save_hidden);
+    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<32, 64>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
   else if (rows == 64 && maxw == 64)
-    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<64, 64>), g, b, 0, stream, args, x, save_hidden);
-  else if (rows == 32 && maxw == 256)
-    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<32, 256>), g, b, 0, stream, args, x, save_hidden);
+    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<64, 64>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
   else
-    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<64, 256>), g, b, 0, stream, args, x, save_hidden);
+    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<32, 256>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
 }
 
 void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
-                          const float* W, float* dx, float* ws, int batch,
-                          int out_d, int in_d, int act, int rows, int maxw,
-                          int n_blocks, hipStream_t stream) {
+                          const float* W, float* dx, float* ws, long ws_stride,
+                          int batch, int out_d, int in_d, int act, int rows,
+                          int maxw, int n_blocks, int wstage_mode,
+                          size_t lds_bytes, hipStream_t stream) {
   dim3 g(n_blocks), b(256);
   if (rows == 32 && maxw == 64)
-    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<32, 64>), g, b, 0, stream, dy, y, xin, W, dx, ws, batch, out_d, in_d, act);
+    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<32, 64>), g, b, lds_bytes, stream, dy, y, xin, W, dx, ws, ws_stride, batch, out_d, in_d, act, wstage_mode);
   else if (rows == 64 && maxw == 64)
-    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<64, 64>), g, b, 0, stream, dy, y, xin, W, dx, ws, batch, out_d, in_d, act);
-  else if (rows == 32 && maxw == 256)
-    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<32, 256>), g, b, 0, stream, dy, y, xin, W, dx, ws, batch, out_d, in_d, act);
+    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<64, 64>), g, b, lds_bytes, stream, dy, y, xin, W, dx, ws, ws_stride, batch, out_d, in_d, act, wstage_mode);
   else
-    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<64, 256>), g, b, 0, stream, dy, y, xin, W, dx, ws, batch, out_d, in_d, act);
+    hipLaunchKernelGGL((mlp_bwd_layer_f32_t<32, 256>), g, b, lds_bytes, stream, dy, y, xin, W, dx, ws, ws_stride, batch, out_d, in_d, act, wstage_mode);
 }
 
-// all-layer deterministic partial reduction: one launch per backward.
-// workspace holds per-layer segments of n_blocks partials each;
-// fixed (s0+s1)+(s2+s3) accumulation order -> bitwise reproducible.
+// ---------------------------------------------------------------------------
+// two-stage deterministic partial reduction
+// stage 1 (layer-blind): out[c][e] = sum_{p in chunk c} ws[p][e]
+// stage 2: dw/db[e] = fixed-order sum over the <=REDUCE_CHUNKS chunk rows
+// ---------------------------------------------------------------------------
+__global__ void mlp_grad_reduce_stage_f32(const float* __restrict__ ws,
+                                          float* __restrict__ out, int n_blocks,
+                                          int chunk, long grand) {
+  const int c = blockIdx.y;
+  const int p0 = c * chunk;
+  const int p1 = min(p0 + chunk, n_blocks);
+  for (long e = blockIdx.x * blockDim.x + threadIdx.x; e < grand;
+       e += (long)gridDim.x * blockDim.x) {
+    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+    int p = p0;
+    for (; p + 3 < p1; p += 4) {
+      s0 += ws[(p + 0) * grand + e];
+      s1 += ws[(p + 1) * grand + e];
+      s2 += ws[(p + 2) * grand + e];
+      s3 += ws[(p + 3) * grand + e];
+    }
+    for (; p < p1; ++p) s0 += ws[p * grand + e];
+    out[c * grand + e] = (s0 + s1) + (s2 + s3);
+  }
+}
+
 __global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a) {
-  // flatten all layers' elements into one grid-stride loop
   int grand = 0;
   int base[MLP_MAX_LAYERS];
   for (int l = 0; l < a.n_layers; ++l) {
@@ -249,18 +355,8 @@ __global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a) {
     int l = 0;
     while (l + 1 < a.n_layers && g >= base[l + 1]) ++l;
     const int idx = g - base[l];
-    const long stride = a.total[l];
-    const float* ws = a.ws[l];
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    int p = 0;
-    for (; p + 3 < a.n_blocks; p += 4) {
-      s0 += ws[(p + 0) * stride + idx];
-      s1 += ws[(p + 1) * stride + idx];
-      s2 += ws[(p + 2) * stride + idx];
-      s3 += ws[(p + 3) * stride + idx];
-    }
-    for (; p < a.n_blocks; ++p) s0 += ws[p * stride + idx];
-    const float s = (s0 + s1) + (s2 + s3);
+    float s = 0.f;
+    for (int p = 0; p < a.n_blocks; ++p) s += a.ws[p * a.stride + g];
     if (idx < a.wsize[l]) a.dw[l][idx] = s;
     else a.db[l][idx - a.wsize[l]] = s;
   }
