@@ -90,6 +90,11 @@ class Experience:
     # ------------------------------------------------------------------
     # MI355X device-pipeline view
     # ------------------------------------------------------------------
+    def set_flat_cache(self, flat: Dict[str, np.ndarray]) -> None:
+        """Install a precomputed flat view (the vectorized sampler builds
+        it in O(1) numpy ops instead of per-episode concatenation)."""
+        self._flat_cache = flat
+
     def to_flat_batch(self) -> Dict[str, np.ndarray]:
         """Contiguous arrays + episode offsets for one-shot H2D upload.
 
@@ -98,6 +103,9 @@ class Experience:
           episode_offsets [N+1] int32 (episode e = rows offsets[e]:offsets[e+1]),
           episode_dones [N] bool, last_observations [N, O*] float32.
         """
+        cached = getattr(self, "_flat_cache", None)
+        if cached is not None:
+            return cached
         lengths = np.asarray(self.episode_lengths, dtype=np.int64)
         offsets = np.zeros(len(lengths) + 1, dtype=np.int32)
         np.cumsum(lengths, out=offsets[1:])

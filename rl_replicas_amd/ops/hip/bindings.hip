@@ -123,6 +123,16 @@ void pick_wstage(const std::vector<torch::Tensor>& weights, int rows, int maxw,
     *mode = 0;
     *lds_bytes = mode0_bytes;
   } else {
+    // wide nets: direct global W reads (mode 2) measure faster than the
+    // serialized per-wave chunk staging (mode 1) at these grid sizes
+    *mode = 2;
+    *lds_bytes = act_elems * 4;
+  }
+  static int env_mode = []() {
+    const char* e = getenv("RL_REPLICAS_AMD_WSTAGE");
+    return e ? atoi(e) : -1;
+  }();
+  if (env_mode == 1 && *mode == 2) {
     *mode = 1;
     size_t wv = backward_single_layer ? (size_t)4 * KCHUNK * 18
                                       : (size_t)4 * 16 * (KCHUNK + 2);

@@ -128,6 +128,15 @@ __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
           float bv = (jok && kk < in_d) ? wl[j * wrow + kk] : 0.f;
           acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
         }
+      } else if (wstage_mode == 2) {
+        // direct global W reads (wide nets at large grids: the L2
+        // misses hide behind cross-wave parallelism)
+        for (int k0 = 0; k0 < in_d; k0 += 4) {
+          const int kk = k0 + k;
+          float a = (kk < in_d) ? buf_in[(wr0 + i) * LDSW + kk] : 0.f;
+          float bv = (jok && kk < in_d) ? W[(long)j * in_d + kk] : 0.f;
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
       } else {
         // per-wave staged K-chunks of W[jt..jt+16)[c0..c0+KCHUNK)
         for (int c0 = 0; c0 < in_d; c0 += KCHUNK) {
@@ -228,6 +237,13 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
         const int kk = k0 + k;
         float a = (kk < out_d) ? dz[(wr0 + i) * LDSW + kk] : 0.f;
         float bv = (jok && kk < out_d) ? wlds[kk * wrow + j] : 0.f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+      }
+    } else if (wstage_mode == 2) {
+      for (int k0 = 0; k0 < out_d; k0 += 4) {
+        const int kk = k0 + k;
+        float a = (kk < out_d) ? dz[(wr0 + i) * LDSW + kk] : 0.f;
+        float bv = (jok && kk < out_d) ? W[(long)kk * in_d + j] : 0.f;
         acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
       }
     } else {

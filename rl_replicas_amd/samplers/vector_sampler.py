@@ -103,6 +103,28 @@ class VectorSampler(Sampler):
                     done_i[start:steps],
                     np.asarray(self.observations[i]),
                 )
+
+        # flat (instance-major, time-ordered) view in O(1) numpy ops —
+        # exactly the order the per-episode emission above produces
+        lengths = np.asarray(experience.episode_lengths, dtype=np.int64)
+        offsets = np.zeros(len(lengths) + 1, dtype=np.int32)
+        np.cumsum(lengths, out=offsets[1:])
+        flat_obs = np.ascontiguousarray(obs_arr.transpose(1, 0, *range(2, obs_arr.ndim))).reshape(
+            steps * n_envs, *obs_arr.shape[2:]
+        )
+        flat_act = np.ascontiguousarray(act_arr.transpose(1, 0, *range(2, act_arr.ndim))).reshape(
+            steps * n_envs, *act_arr.shape[2:]
+        )
+        experience.set_flat_cache(
+            {
+                "observations": flat_obs.astype(np.float32, copy=False),
+                "actions": flat_act,
+                "rewards": np.ascontiguousarray(rew_arr.T).reshape(-1).astype(np.float32),
+                "episode_offsets": offsets,
+                "episode_dones": np.asarray(experience.episode_dones, dtype=bool),
+                "last_observations": np.stack(experience.last_observations).astype(np.float32),
+            }
+        )
         return experience
 
     @staticmethod

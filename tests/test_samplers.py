@@ -139,3 +139,20 @@ class TestVectorSampler:
         sampler = VectorSampler(venv, seed=0)
         with pytest.raises(ValueError):
             sampler.sample(100, RandomPolicy(venv.action_space))
+
+
+def test_vector_sampler_flat_cache_matches_recompute():
+    """The O(1) flat view the sampler caches equals the per-episode
+    concatenation path."""
+    venv = envs.VectorEnv("CartPole-v1", num_envs=5)
+    venv.action_space.seed(3)
+    s = VectorSampler(venv, seed=3)
+    exp = s.sample(200, RandomPolicy(venv.action_space))
+    cached = exp.to_flat_batch()
+    exp._flat_cache = None
+    recomputed = exp.to_flat_batch()
+    for key in recomputed:
+        np.testing.assert_array_equal(
+            np.asarray(cached[key], dtype=np.asarray(recomputed[key]).dtype),
+            recomputed[key],
+        )
