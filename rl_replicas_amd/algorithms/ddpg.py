@@ -57,23 +57,43 @@ class DDPG(OffPolicyAlgorithm):
 
     # ------------------------------------------------------------------
     def train(self, replay_buffer: ReplayBuffer, num_train_steps: int, minibatch_size: int) -> None:
-        policy_losses: List[float] = []
-        q_losses: List[float] = []
+        """50-iteration minibatch loop (reference ddpg.py:195-253).
+
+        Losses/Q-values stay device-resident across the loop (one
+        readback at the end); on GPU the Q and actor steps run through
+        the fused kernel path (ops.fused_offpolicy)."""
+        from rl_replicas_amd.ops import fused_offpolicy as fop
+
+        policy_losses: List[Tensor] = []
+        q_losses: List[Tensor] = []
         all_q_values: List[Tensor] = []
 
         for _ in range(num_train_steps):
             mb = self._sample_minibatch_device(minibatch_size)
             observations = mb["observations"]
             actions = mb["actions"]
+            fused = fop.supported(self.q_function, observations) and fop.supported(
+                self.policy, observations
+            )
 
             with torch.no_grad():
                 all_q_values.append(self.q_function(observations, actions))
 
             targets = self.compute_targets(mb["next_observations"], mb["rewards"], mb["dones"])
-            q_losses.append(
-                self._train_q_single(self.q_function, observations, actions, targets).item()
-            )
-            policy_losses.append(self.train_policy(observations).item())
+            if fused:
+                q_losses.append(
+                    fop.q_step(self.q_function, observations, actions, targets,
+                               self._all_reduce_gradients)
+                )
+                policy_losses.append(
+                    fop.policy_step(self.policy, self.q_function, observations,
+                                    self._all_reduce_gradients)
+                )
+            else:
+                q_losses.append(
+                    self._train_q_single(self.q_function, observations, actions, targets)
+                )
+                policy_losses.append(self.train_policy(observations))
 
             polyak_average(
                 self.policy.network.parameters(),
@@ -89,10 +109,16 @@ class DDPG(OffPolicyAlgorithm):
         q_values = torch.cat(all_q_values)
         m = self.metrics_manager
         m.record_scalar(
-            "policy/average_loss", float(np.mean(policy_losses)), self.current_total_steps, tensorboard=True
+            "policy/average_loss",
+            float(torch.stack(policy_losses).mean()),
+            self.current_total_steps,
+            tensorboard=True,
         )
         m.record_scalar(
-            "q-function/average_loss", float(np.mean(q_losses)), self.current_total_steps, tensorboard=True
+            "q-function/average_loss",
+            float(torch.stack(q_losses).mean()),
+            self.current_total_steps,
+            tensorboard=True,
         )
         m.record_scalar(
             "q-function/avarage_q-value", float(q_values.mean()), self.current_total_steps, tensorboard=True

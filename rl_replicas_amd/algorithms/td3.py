@@ -66,9 +66,13 @@ class TD3(OffPolicyAlgorithm):
 
     # ------------------------------------------------------------------
     def train(self, replay_buffer: ReplayBuffer, num_train_steps: int, minibatch_size: int) -> None:
-        policy_losses: List[float] = []
-        q1_losses: List[float] = []
-        q2_losses: List[float] = []
+        """Twin-critic minibatch loop (reference td3.py:214-263); losses
+        stay device-resident, fused kernel steps on GPU."""
+        from rl_replicas_amd.ops import fused_offpolicy as fop
+
+        policy_losses: List[Tensor] = []
+        q1_losses: List[Tensor] = []
+        q2_losses: List[Tensor] = []
         all_q1: List[Tensor] = []
         all_q2: List[Tensor] = []
 
@@ -76,21 +80,40 @@ class TD3(OffPolicyAlgorithm):
             mb = self._sample_minibatch_device(minibatch_size)
             observations = mb["observations"]
             actions = mb["actions"]
+            fused = fop.supported(self.q_function_1, observations) and fop.supported(
+                self.policy, observations
+            )
 
             with torch.no_grad():
                 all_q1.append(self.q_function_1(observations, actions))
                 all_q2.append(self.q_function_2(observations, actions))
 
             targets = self.compute_targets(mb["next_observations"], mb["rewards"], mb["dones"])
-            q1_losses.append(
-                self._train_q_single(self.q_function_1, observations, actions, targets).item()
-            )
-            q2_losses.append(
-                self._train_q_single(self.q_function_2, observations, actions, targets).item()
-            )
+            if fused:
+                q1_losses.append(
+                    fop.q_step(self.q_function_1, observations, actions, targets,
+                               self._all_reduce_gradients)
+                )
+                q2_losses.append(
+                    fop.q_step(self.q_function_2, observations, actions, targets,
+                               self._all_reduce_gradients)
+                )
+            else:
+                q1_losses.append(
+                    self._train_q_single(self.q_function_1, observations, actions, targets)
+                )
+                q2_losses.append(
+                    self._train_q_single(self.q_function_2, observations, actions, targets)
+                )
 
             if train_step % self.policy_delay == 0:
-                policy_losses.append(self.train_policy(observations).item())
+                if fused:
+                    policy_losses.append(
+                        fop.policy_step(self.policy, self.q_function_1, observations,
+                                        self._all_reduce_gradients)
+                    )
+                else:
+                    policy_losses.append(self.train_policy(observations))
                 polyak_average(
                     self.policy.network.parameters(),
                     self.target_policy.network.parameters(),
@@ -111,13 +134,22 @@ class TD3(OffPolicyAlgorithm):
         q2 = torch.cat(all_q2)
         m = self.metrics_manager
         m.record_scalar(
-            "policy/average_loss", float(np.mean(policy_losses)), self.current_total_steps, tensorboard=True
+            "policy/average_loss",
+            float(torch.stack(policy_losses).mean()),
+            self.current_total_steps,
+            tensorboard=True,
         )
         m.record_scalar(
-            "q-function_1/average_loss", float(np.mean(q1_losses)), self.current_total_steps, tensorboard=True
+            "q-function_1/average_loss",
+            float(torch.stack(q1_losses).mean()),
+            self.current_total_steps,
+            tensorboard=True,
         )
         m.record_scalar(
-            "q-function_2/average_loss", float(np.mean(q2_losses)), self.current_total_steps, tensorboard=True
+            "q-function_2/average_loss",
+            float(torch.stack(q2_losses).mean()),
+            self.current_total_steps,
+            tensorboard=True,
         )
         m.record_scalar(
             "q-function_1/avarage_q-value", float(q1.mean()), self.current_total_steps, tensorboard=True
