@@ -1,0 +1,28 @@
+"""VPG benchmark run (reference config: benchmarks/run_vpg.py:28-55 —
+750 epochs x 4000 steps = 3M env steps)."""
+import argparse
+import os
+import sys
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
+from common import build_on_policy  # noqa: E402
+
+
+def run_vpg(env_id, seed, outdir, device=None, num_envs=20, num_epochs=None):
+    from rl_replicas_amd.algorithms import VPG
+
+    env, sampler, policy, value_function = build_on_policy(env_id, seed, device, num_envs, "adam")
+    model = VPG(policy, value_function, env, sampler)
+    model.learn(num_epochs=num_epochs or 750, batch_size=4000, output_dir=outdir)
+
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("--env", default="HalfCheetah-v4")
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--outdir", default=".")
+    p.add_argument("--device", default=None)
+    p.add_argument("--num-envs", type=int, default=20)
+    p.add_argument("--num-epochs", type=int, default=None)
+    a = p.parse_args()
+    run_vpg(a.env, a.seed, a.outdir, a.device, a.num_envs, a.num_epochs)

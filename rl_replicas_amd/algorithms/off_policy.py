@@ -81,6 +81,8 @@ class OffPolicyAlgorithm(AlgorithmBase):
         model_saving_interval: int = 4000,
         output_dir: str = ".",
     ) -> None:
+        import time as _time
+
         self._begin_learn(output_dir)
         for current_epoch in range(1, num_epochs + 1):
             behavior = (
@@ -88,8 +90,10 @@ class OffPolicyAlgorithm(AlgorithmBase):
                 if self.current_total_steps < num_start_steps
                 else self.noised_policy
             )
+            t0 = _time.perf_counter()
             experience: Experience = self.sampler.sample(batch_size, behavior)
             self.replay_buffer.add_experience(experience)
+            t1 = _time.perf_counter()
 
             self.current_total_steps += sum(experience.episode_lengths)
             self.current_total_episodes += sum(experience.flattened_dones)
@@ -100,12 +104,17 @@ class OffPolicyAlgorithm(AlgorithmBase):
 
             if self.current_total_steps >= num_steps_before_update:
                 self.train(self.replay_buffer, num_train_steps, minibatch_size)
+            self.metrics_manager.record_phase_ms("sample", (t1 - t0) * 1000.0)
+            self.metrics_manager.record_phase_ms("train", (_time.perf_counter() - t1) * 1000.0)
 
             if (
                 num_evaluation_episodes > 0
                 and self.current_total_steps % evaluation_interval == 0
             ):
+                t2 = _time.perf_counter()
                 self._run_evaluation(num_evaluation_episodes)
+                self.metrics_manager.record_phase_ms("evaluate", (_time.perf_counter() - t2) * 1000.0)
+            self.metrics_manager.dump_phases(self.current_total_steps)
 
             self._end_epoch(current_epoch, model_saving_interval, output_dir)
         self.metrics_manager.close()

@@ -72,9 +72,13 @@ class OnPolicyAlgorithm(AlgorithmBase):
 
         Reference: e.g. ppo.py:72-137.
         """
+        import time as _time
+
         self._begin_learn(output_dir)
         for current_epoch in range(1, num_epochs + 1):
+            t0 = _time.perf_counter()
             experience = self.sampler.sample(batch_size, self.policy)
+            t1 = _time.perf_counter()
 
             self.current_total_steps += sum(experience.episode_lengths)
             self.current_total_episodes += sum(experience.episode_dones)
@@ -84,6 +88,11 @@ class OnPolicyAlgorithm(AlgorithmBase):
             )
 
             self.train(experience)
+
+            # per-phase observability the reference lacks
+            self.metrics_manager.record_phase_ms("sample", (t1 - t0) * 1000.0)
+            self.metrics_manager.record_phase_ms("train", (_time.perf_counter() - t1) * 1000.0)
+            self.metrics_manager.dump_phases(self.current_total_steps)
 
             self._end_epoch(current_epoch, model_saving_interval, output_dir)
         self.metrics_manager.close()
