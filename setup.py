@@ -37,6 +37,7 @@ setup(
     name="rl_replicas_amd",
     version="0.1.0",
     packages=[
+        "rl_replicas",
         "rl_replicas_amd",
         "rl_replicas_amd.algorithms",
         "rl_replicas_amd.envs",

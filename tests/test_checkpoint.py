@@ -119,3 +119,18 @@ def test_reference_checkpoint_loads():
     model.policy.network.load_state_dict(ref_policy_sd)
     for k in ref_policy_sd:
         torch.testing.assert_close(model.policy.network.state_dict()[k], ref_policy_sd[k])
+
+
+def test_rl_replicas_drop_in_alias():
+    """Reference-style imports work through the rl_replicas shim."""
+    from rl_replicas.algorithms import PPO as AliasPPO
+    from rl_replicas.policies import CategoricalPolicy as AliasCat
+    from rl_replicas_amd.algorithms import PPO
+
+    assert AliasPPO is PPO
+    from rl_replicas.utils import discounted_cumulative_sums
+    import numpy as np
+
+    np.testing.assert_allclose(
+        discounted_cumulative_sums(np.array([1.0, 1.0]), 0.5), [1.5, 1.0]
+    )
