@@ -55,6 +55,11 @@ def main() -> None:
                 out = os.path.join(args.outdir, env_id, algorithm, f"seed-{seed}")
                 os.makedirs(out, exist_ok=True)
                 print(f"=== {algorithm} / {env_id} / seed {seed} -> {out}")
+                # off-policy epochs sample batch_size=50 steps: the env
+                # count must divide it
+                num_envs = args.num_envs
+                if algorithm in ("ddpg", "td3"):
+                    num_envs = max(d for d in (1, 2, 5, 10, 25, 50) if d <= num_envs)
                 # stdout -> experiment.log (reference run_vpg.py:49-56)
                 with open(os.path.join(out, "experiment.log"), "w") as log:
                     with contextlib.redirect_stdout(log):
@@ -63,7 +68,7 @@ def main() -> None:
                             seed,
                             out,
                             device=args.device,
-                            num_envs=args.num_envs,
+                            num_envs=num_envs,
                             num_epochs=args.num_epochs,
                         )
 
