@@ -22,6 +22,13 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
                           int batch, int out_d, int in_d, int act, int rows,
                           int maxw, int n_blocks, int wstage_mode,
                           size_t lds_bytes, hipStream_t stream);
+void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
+                               float* out, int batch, int in_d, int out_d,
+                               int act, hipStream_t stream);
+void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
+                         const float* W, float* dx, float* ws, long ws_stride,
+                         int batch, int out_d, int in_d, int act,
+                         hipStream_t stream);
 __global__ void mlp_grad_reduce_stage_f32(const float* ws, float* out,
                                           int n_blocks, int chunk, long grand);
 __global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a);
@@ -171,33 +178,50 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
     args.acts[l] = (int)acts[l];
   }
 
+  int max_width = args.dims[0];
+  for (int l = 1; l <= L; ++l) max_width = std::max(max_width, args.dims[l]);
+  int rows, maxw;
+  pick_tile(args.batch, max_width, &rows, &maxw);
+
   auto opts = x.options();
+  const bool wide = maxw == 256;
   std::vector<torch::Tensor> outs;  // [final, h0..h_{L-2}]
   torch::Tensor final_out = torch::empty({x.size(0), args.dims[L]}, opts);
   outs.push_back(final_out);
+  std::vector<torch::Tensor> inter;  // wide path needs intermediates always
   for (int l = 0; l < L - 1; ++l) {
-    if (save_hidden) {
+    if (save_hidden || wide) {
       torch::Tensor h = torch::empty({x.size(0), args.dims[l + 1]}, opts);
       args.h[l] = h.data_ptr<float>();
-      outs.push_back(h);
+      if (save_hidden) outs.push_back(h);
+      else inter.push_back(h);
     } else {
       args.h[l] = nullptr;
     }
   }
   args.h[L - 1] = final_out.data_ptr<float>();
 
-  int max_width = args.dims[0];
-  for (int l = 1; l <= L; ++l) max_width = std::max(max_width, args.dims[l]);
-  int rows, maxw;
-  pick_tile(args.batch, max_width, &rows, &maxw);
-  const int n_blocks = (args.batch + rows - 1) / rows;
-  int wmode;
-  size_t lds_bytes;
-  pick_wstage(weights, rows, maxw, false, 0, &wmode, &lds_bytes);
-  if (n_blocks > 0) {
-    launch_mlp_fwd(args, x.data_ptr<float>(), save_hidden ? 1 : 0, rows, maxw,
-                   n_blocks, wmode, lds_bytes, current_stream());
-    HIP_OK(hipGetLastError());
+  if (args.batch > 0) {
+    auto stream = current_stream();
+    if (wide) {
+      // per-layer 2D-grid kernels: column groups spread across blocks
+      const float* cur = x.data_ptr<float>();
+      for (int l = 0; l < L; ++l) {
+        launch_mlp_layer_fwd_wide(cur, args.w[l], args.b[l], args.h[l],
+                                  args.batch, args.dims[l], args.dims[l + 1],
+                                  args.acts[l], stream);
+        HIP_OK(hipGetLastError());
+        cur = args.h[l];
+      }
+    } else {
+      const int n_blocks = (args.batch + rows - 1) / rows;
+      int wmode;
+      size_t lds_bytes;
+      pick_wstage(weights, rows, maxw, false, 0, &wmode, &lds_bytes);
+      launch_mlp_fwd(args, x.data_ptr<float>(), save_hidden ? 1 : 0, rows, maxw,
+                     n_blocks, wmode, lds_bytes, stream);
+      HIP_OK(hipGetLastError());
+    }
   }
   return outs;
 }
@@ -245,15 +269,23 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dws[l] = torch::empty({out_d, in_d}, opts);
     dbs[l] = torch::empty({out_d}, opts);
     dx = torch::empty({batch, in_d}, opts);
-    int wmode;
-    size_t lds_bytes;
-    pick_wstage(weights, rows, maxw, true, l, &wmode, &lds_bytes);
-    // merged dgrad + wgrad/bias partials in one kernel
-    launch_mlp_bwd_layer(dy.data_ptr<float>(), y.data_ptr<float>(),
-                         xin.data_ptr<float>(), weights[l].data_ptr<float>(),
-                         dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
-                         batch, out_d, in_d, (int)acts[l], rows, maxw, n_blocks,
-                         wmode, lds_bytes, stream);
+    if (maxw == 256) {
+      // wide layer: 2D-grid dgrad + wgrad kernels
+      launch_mlp_bwd_wide(dy.data_ptr<float>(), y.data_ptr<float>(),
+                          xin.data_ptr<float>(), weights[l].data_ptr<float>(),
+                          dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
+                          batch, out_d, in_d, (int)acts[l], stream);
+    } else {
+      int wmode;
+      size_t lds_bytes;
+      pick_wstage(weights, rows, maxw, true, l, &wmode, &lds_bytes);
+      // merged dgrad + wgrad/bias partials in one kernel
+      launch_mlp_bwd_layer(dy.data_ptr<float>(), y.data_ptr<float>(),
+                           xin.data_ptr<float>(), weights[l].data_ptr<float>(),
+                           dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
+                           batch, out_d, in_d, (int)acts[l], rows, maxw,
+                           n_blocks, wmode, lds_bytes, stream);
+    }
     HIP_OK(hipGetLastError());
     dy = dx;
   }

@@ -305,8 +305,192 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
   }
 }
 
+// ---------------------------------------------------------------------------
+// wide-net single-layer kernels: 2D grids (row blocks x 64-col groups).
+// The fused multi-layer kernel keeps every output column of a row tile
+// in one block, which leaves most of the chip idle for [*,256,256,*]
+// nets at minibatch-sized inputs (DDPG/TD3's 100-row hot path); these
+// per-layer kernels spread the column dimension across blocks instead.
+// ---------------------------------------------------------------------------
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_layer_fwd_wide_f32(
+    const float* __restrict__ x, const float* __restrict__ W,
+    const float* __restrict__ B, float* __restrict__ out, int batch, int in_d,
+    int out_d, int act) {
+  constexpr int LDSW = 256 + 4;
+  extern __shared__ float smem[];  // x tile [ROWS][LDSW]
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = blockIdx.x * ROWS;
+  const int jt = blockIdx.y * 64 + wave * 16;  // one 16-col tile per wave
+
+  load_tile<LDSW>(x, smem, row0, batch, in_d, tid, ROWS);
+  __syncthreads();
+  if (jt >= out_d) return;
+
+  const int i = lane & 15;
+  const int k = lane >> 4;
+  const int j = jt + i;
+  const bool jok = j < out_d;
+  const float bias = jok ? B[j] : 0.f;
+  for (int rt = 0; rt < ROWS; rt += 16) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < in_d; k0 += 4) {
+      const int kk = k0 + k;
+      float a = (kk < in_d) ? smem[(rt + i) * LDSW + kk] : 0.f;
+      float bv = (jok && kk < in_d) ? W[(long)j * in_d + kk] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    }
+    if (jok) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + rt + (lane >> 4) * 4 + r;
+        if (row < batch) out[(long)row * out_d + j] = act_apply(act, acc[r] + bias);
+      }
+    }
+  }
+}
+
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_dgrad_wide_f32(
+    const float* __restrict__ dy, const float* __restrict__ y,
+    const float* __restrict__ W, float* __restrict__ dx, int batch, int out_d,
+    int in_d, int act) {
+  constexpr int LDSW = 256 + 4;
+  extern __shared__ float smem[];  // dz tile [ROWS][LDSW]
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = blockIdx.x * ROWS;
+  const int jt = blockIdx.y * 64 + wave * 16;  // over in_d
+
+  for (int idx = tid; idx < ROWS * out_d; idx += 256) {
+    int r = idx / out_d, c = idx % out_d;
+    int row = row0 + r;
+    float v = 0.f;
+    if (row < batch) {
+      long g = (long)row * out_d + c;
+      v = dy[g] * act_grad_from_y(act, y[g]);
+    }
+    smem[r * LDSW + c] = v;
+  }
+  __syncthreads();
+  if (jt >= in_d) return;
+
+  const int i = lane & 15;
+  const int k = lane >> 4;
+  const int j = jt + i;
+  const bool jok = j < in_d;
+  for (int rt = 0; rt < ROWS; rt += 16) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < out_d; k0 += 4) {
+      const int kk = k0 + k;
+      float a = (kk < out_d) ? smem[(rt + i) * LDSW + kk] : 0.f;
+      float bv = (jok && kk < out_d) ? W[(long)kk * in_d + j] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    }
+    if (jok) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = row0 + rt + (lane >> 4) * 4 + r;
+        if (row < batch) dx[(long)row * in_d + j] = acc[r];
+      }
+    }
+  }
+}
+
+// wgrad + bias partials, columns of dW's out dimension split over
+// blockIdx.y (disjoint writes into the same per-row-block partial row)
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
+    const float* __restrict__ dy, const float* __restrict__ y,
+    const float* __restrict__ xin, float* __restrict__ workspace,
+    long ws_stride, int batch, int out_d, int in_d, int act) {
+  constexpr int LDSW = 256 + 4;
+  constexpr int DZW = 64 + 4;
+  extern __shared__ float smem[];  // xt [ROWS][LDSW] + dz slice [ROWS][DZW]
+  float* xt = smem;
+  float* dz = smem + ROWS * LDSW;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = blockIdx.x * ROWS;
+  const int og0 = blockIdx.y * 64;  // out-col group
+  float* wsp = workspace + (long)blockIdx.x * ws_stride;
+
+  load_tile<LDSW>(xin, xt, row0, batch, in_d, tid, ROWS);
+  const int og_w = min(64, out_d - og0);
+  for (int idx = tid; idx < ROWS * og_w; idx += 256) {
+    int r = idx / og_w, c = idx % og_w;
+    int row = row0 + r;
+    float v = 0.f;
+    if (row < batch) {
+      long g = (long)row * out_d + og0 + c;
+      v = dy[g] * act_grad_from_y(act, y[g]);
+    }
+    dz[r * DZW + c] = v;
+  }
+  __syncthreads();
+
+  const int i = lane & 15;
+  const int k = lane >> 4;
+  const int ii = wave * 16 + i;       // dz slice col (out index og0+ii)
+  const bool iok = og0 + ii < out_d;
+  for (int jt = 0; jt < in_d; jt += 16) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < ROWS; k0 += 4) {
+      float a = iok ? dz[(k0 + k) * DZW + ii] : 0.f;
+      float bv = (jt + i < in_d) ? xt[(k0 + k) * LDSW + jt + i] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    }
+    const int col = jt + i;
+    if (col < in_d) {
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int orow = og0 + wave * 16 + (lane >> 4) * 4 + r;
+        if (orow < out_d) wsp[(long)orow * in_d + col] = acc[r];
+      }
+    }
+  }
+
+  __syncthreads();
+  for (int c = tid; c < og_w; c += 256) {
+    float s = 0.f;
+    #pragma unroll 4
+    for (int r = 0; r < ROWS; ++r) s += dz[r * DZW + c];
+    wsp[(long)out_d * in_d + og0 + c] = s;
+  }
+}
+
 // host-side dispatch over the (ROWS, MAXW) instantiations — called from
 // bindings.hip so template symbols stay in this translation unit
+void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
+                               float* out, int batch, int in_d, int out_d,
+                               int act, hipStream_t stream) {
+  constexpr int ROWS = 32;
+  dim3 g((batch + ROWS - 1) / ROWS, (out_d + 63) / 64), b(256);
+  size_t lds = (size_t)ROWS * (256 + 4) * 4;
+  hipLaunchKernelGGL((mlp_layer_fwd_wide_f32<ROWS>), g, b, lds, stream, x, W, B,
+                     out, batch, in_d, out_d, act);
+}
+
+void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
+                         const float* W, float* dx, float* ws, long ws_stride,
+                         int batch, int out_d, int in_d, int act,
+                         hipStream_t stream) {
+  constexpr int ROWS = 32;
+  const int rb = (batch + ROWS - 1) / ROWS;
+  size_t lds1 = (size_t)ROWS * (256 + 4) * 4;
+  hipLaunchKernelGGL((mlp_dgrad_wide_f32<ROWS>), dim3(rb, (in_d + 63) / 64),
+                     dim3(256), lds1, stream, dy, y, W, dx, batch, out_d, in_d,
+                     act);
+  size_t lds2 = (size_t)ROWS * (256 + 4 + 64 + 4) * 4;
+  hipLaunchKernelGGL((mlp_wgrad_wide_f32<ROWS>), dim3(rb, (out_d + 63) / 64),
+                     dim3(256), lds2, stream, dy, y, xin, ws, ws_stride, batch,
+                     out_d, in_d, act);
+}
+
 void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
                     int rows, int maxw, int n_blocks, int wstage_mode,
                     size_t lds_bytes, hipStream_t stream) {
