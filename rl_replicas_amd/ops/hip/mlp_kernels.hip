@@ -334,12 +334,29 @@ __global__ __launch_bounds__(256) void mlp_layer_fwd_wide_f32(
   const int j = jt + i;
   const bool jok = j < out_d;
   const float bias = jok ? B[j] : 0.f;
+  const float* wrow = W + (long)j * in_d;
+  const int in4 = in_d & ~3;
   for (int rt = 0; rt < ROWS; rt += 16) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int k0 = 0; k0 < in_d; k0 += 4) {
-      const int kk = k0 + k;
+    // every lane of a 16-lane group loads the same 4-float W chunk
+    // (hardware broadcast) and selects its K element: 4x fewer load
+    // instructions than per-step scalar reads, fully pipelineable
+    for (int k0 = 0; k0 < in4; k0 += 4) {
+      float w0 = 0.f, w1 = 0.f, w2 = 0.f, w3 = 0.f;
+      if (jok) {
+        w0 = wrow[k0 + 0];
+        w1 = wrow[k0 + 1];
+        w2 = wrow[k0 + 2];
+        w3 = wrow[k0 + 3];
+      }
+      const float bv = k == 0 ? w0 : (k == 1 ? w1 : (k == 2 ? w2 : w3));
+      const float a = smem[(rt + i) * LDSW + k0 + k];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+    }
+    if (in4 < in_d) {  // ragged K tail
+      const int kk = in4 + k;
       float a = (kk < in_d) ? smem[(rt + i) * LDSW + kk] : 0.f;
-      float bv = (jok && kk < in_d) ? W[(long)j * in_d + kk] : 0.f;
+      float bv = (jok && kk < in_d) ? wrow[kk] : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
     }
     if (jok) {
