@@ -22,6 +22,9 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
                           int batch, int out_d, int in_d, int act, int rows,
                           int maxw, int n_blocks, int wstage_mode,
                           size_t lds_bytes, hipStream_t stream);
+void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
+                          const float* dy, float* dx, float* ws,
+                          size_t lds_bytes, int n_blocks, hipStream_t stream);
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
                                float* out, int batch, int in_d, int out_d,
                                int act, hipStream_t stream);
@@ -255,6 +258,72 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     layer_off[l] = grand;
     grand += totals[l];
   }
+  // whole-net fused backward for narrow nets when the LDS image fits
+  size_t whole_w = 0;
+  for (auto& w : weights) whole_w += (size_t)w.size(0) * (w.size(1) + 1);
+  const size_t fused_lds = ((size_t)3 * 32 * 68 + whole_w) * 4;
+  const bool use_fused_bwd = (maxw == 64) && fused_lds <= 100 * 1024;
+  if (use_fused_bwd) {
+    const int fb = (batch + 31) / 32;
+    torch::Tensor ws = torch::empty({(int64_t)fb, grand}, opts);
+    std::vector<torch::Tensor> dws(L), dbs(L);
+    MLPBwdArgs ba{};
+    ba.n_layers = L;
+    ba.batch = batch;
+    ba.ws_stride = grand;
+    ba.dims[0] = (int)x.size(1);
+    for (int l = 0; l < L; ++l) {
+      ba.w[l] = weights[l].data_ptr<float>();
+      ba.h[l] = (l == L - 1 ? final_out : hidden[l]).data_ptr<float>();
+      ba.dims[l + 1] = (int)weights[l].size(0);
+      ba.acts[l] = (int)acts[l];
+      ba.layer_off[l] = (int)layer_off[l];
+      dws[l] = torch::empty({(int)weights[l].size(0), (int)weights[l].size(1)}, opts);
+      dbs[l] = torch::empty({(int)weights[l].size(0)}, opts);
+    }
+    torch::Tensor dx = torch::empty({batch, x.size(1)}, opts);
+    launch_mlp_bwd_fused(ba, x.data_ptr<float>(),
+                         grad_out.contiguous().data_ptr<float>(),
+                         dx.data_ptr<float>(), ws.data_ptr<float>(), fused_lds,
+                         fb, stream);
+    HIP_OK(hipGetLastError());
+
+    const float* red_src = ws.data_ptr<float>();
+    int red_n = fb;
+    torch::Tensor ws2;
+    if (fb > 16) {
+      const int chunks = 8;
+      const int chunk = (fb + chunks - 1) / chunks;
+      ws2 = torch::empty({(int64_t)chunks, grand}, opts);
+      dim3 g((unsigned)std::min<int64_t>(128, (grand + 255) / 256), chunks);
+      hipLaunchKernelGGL(mlp_grad_reduce_stage_f32, g, dim3(256), 0, stream,
+                         ws.data_ptr<float>(), ws2.data_ptr<float>(), fb, chunk,
+                         grand);
+      HIP_OK(hipGetLastError());
+      red_src = ws2.data_ptr<float>();
+      red_n = chunks;
+    }
+    ReduceAllArgs ra{};
+    ra.ws = red_src;
+    ra.stride = grand;
+    ra.n_layers = L;
+    ra.n_blocks = red_n;
+    for (int l = 0; l < L; ++l) {
+      ra.dw[l] = dws[l].data_ptr<float>();
+      ra.db[l] = dbs[l].data_ptr<float>();
+      ra.total[l] = (int)totals[l];
+      ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+    }
+    int rb = (int)std::min<int64_t>(512, (grand + 255) / 256);
+    hipLaunchKernelGGL(mlp_grad_reduce_all_f32, dim3(rb), dim3(256), 0, stream, ra);
+    HIP_OK(hipGetLastError());
+    std::vector<torch::Tensor> out;
+    out.push_back(dx);
+    for (int l = 0; l < L; ++l) out.push_back(dws[l]);
+    for (int l = 0; l < L; ++l) out.push_back(dbs[l]);
+    return out;
+  }
+
   torch::Tensor ws = torch::empty({(int64_t)n_blocks, grand}, opts);
   float* ws_ptr = ws.data_ptr<float>();
 

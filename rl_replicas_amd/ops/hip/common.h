@@ -56,6 +56,18 @@ struct MLPArgs {
   int batch;
 };
 
+// whole-net fused-backward argument block (mlp_kernels.hip)
+struct MLPBwdArgs {
+  const float* w[MLP_MAX_LAYERS];
+  const float* h[MLP_MAX_LAYERS];  // post-activations; h[L-1] = final out
+  int dims[MLP_MAX_LAYERS + 1];
+  int acts[MLP_MAX_LAYERS];
+  int n_layers;
+  int batch;
+  long ws_stride;
+  int layer_off[MLP_MAX_LAYERS];  // flat elem offset of each layer's partials
+};
+
 // all-layer gradient-reduction argument block (mlp_kernels.hip).
 // workspace layout: ws[partial][flat-elem] with the layers' (od*id+od)
 // segments concatenated in layer order inside each partial row.

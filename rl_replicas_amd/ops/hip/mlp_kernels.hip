@@ -306,6 +306,144 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
 }
 
 // ---------------------------------------------------------------------------
+// whole-net fused backward (narrow nets): ONE kernel walks the layer
+// chain backwards per 32-row block — dZ ping-pongs between two LDS
+// tiles (the next layer's activation-grad is fused into the dgrad
+// epilogue, read from the already-staged X tile), weights come from the
+// whole-net LDS image, and wgrad/bias partials stream to the split-K
+// workspace.  Replaces L per-layer launches with one.
+// ---------------------------------------------------------------------------
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
+    MLPBwdArgs args, const float* __restrict__ x, const float* __restrict__ dy,
+    float* __restrict__ dx_out, float* __restrict__ workspace) {
+  constexpr int MAXW = 64;
+  constexpr int LDSW = MAXW + 4;
+  constexpr int RT = ROWS / 16;
+  constexpr int JT_STRIDE = 4 / RT;
+  extern __shared__ float smem[];
+  float* const dza = smem;
+  float* const dzb = smem + ROWS * LDSW;
+  float* const xt = smem + 2 * ROWS * LDSW;
+  float* const wlds = smem + 3 * ROWS * LDSW;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int row0 = blockIdx.x * ROWS;
+  const int wr0 = (wave % RT) * 16;
+  const int jt0 = (wave / RT) * 16;
+  float* wsp = workspace + (long)blockIdx.x * args.ws_stride;
+  const int L = args.n_layers;
+
+  // whole-net padded W image + layer offsets
+  int woffs[MLP_MAX_LAYERS];
+  {
+    int off = 0;
+    for (int l = 0; l < L; ++l) {
+      woffs[l] = off;
+      stage_weights_block(args.w[l], wlds + off, args.dims[l + 1], args.dims[l], tid);
+      off += args.dims[l + 1] * (args.dims[l] + 1);
+    }
+  }
+  // seed dZ for the last layer: dY * act'(final out)
+  {
+    const int od = args.dims[L];
+    const int act = args.acts[L - 1];
+    const float* yl = args.h[L - 1];
+    for (int idx = tid; idx < ROWS * od; idx += 256) {
+      int r = idx / od, c = idx % od;
+      int row = row0 + r;
+      float v = 0.f;
+      if (row < args.batch) {
+        long g = (long)row * od + c;
+        v = dy[g] * act_grad_from_y(act, yl[g]);
+      }
+      dza[r * LDSW + c] = v;
+    }
+  }
+  __syncthreads();
+
+  const int i = lane & 15;
+  const int k = lane >> 4;
+  const float* dz_cur = dza;
+  float* dz_nxt = dzb;
+
+  for (int l = L - 1; l >= 0; --l) {
+    const int in_d = args.dims[l];
+    const int out_d = args.dims[l + 1];
+    const int wrow = in_d + 1;
+    const float* wl = wlds + woffs[l];
+
+    // stage X_l (input activations of layer l; post-act of layer l-1)
+    load_tile<LDSW>(l == 0 ? x : args.h[l - 1], xt, row0, args.batch, in_d, tid, ROWS);
+    __syncthreads();
+
+    // ---- wgrad partials: dW[i][j] = sum_r dZ[r][i] X[r][j] ----
+    const int n_it = (out_d + 15) / 16;
+    float* lw = wsp + args.layer_off[l];
+    for (int it = wave; it < n_it; it += 4) {
+      const int ii = it * 16 + i;
+      for (int jt = 0; jt < in_d; jt += 16) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        for (int k0 = 0; k0 < ROWS; k0 += 4) {
+          float a = (ii < out_d) ? dz_cur[(k0 + k) * LDSW + ii] : 0.f;
+          float bv = (jt + i < in_d) ? xt[(k0 + k) * LDSW + jt + i] : 0.f;
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
+        const int col = jt + i;
+        if (col < in_d) {
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int orow = it * 16 + (lane >> 4) * 4 + r;
+            if (orow < out_d) lw[(long)orow * in_d + col] = acc[r];
+          }
+        }
+      }
+    }
+    // bias partial
+    for (int c = tid; c < out_d; c += 256) {
+      float s = 0.f;
+      #pragma unroll 4
+      for (int r = 0; r < ROWS; ++r) s += dz_cur[r * LDSW + c];
+      lw[(long)out_d * in_d + c] = s;
+    }
+
+    // ---- dgrad: dX[b][j] = sum_k dZ[b][k] W[k][j]; for l>0 the next
+    // activation-grad is fused from the staged X tile ----
+    const int prev_act = l > 0 ? args.acts[l - 1] : ACT_IDENTITY;
+    for (int jt = jt0; jt < in_d; jt += 16 * JT_STRIDE) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      const int j = jt + i;
+      const bool jok = j < in_d;
+      for (int k0 = 0; k0 < out_d; k0 += 4) {
+        const int kk = k0 + k;
+        float a = (kk < out_d) ? dz_cur[(wr0 + i) * LDSW + kk] : 0.f;
+        float bv = (jok && kk < out_d) ? wl[kk * wrow + j] : 0.f;
+        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+      }
+      if (jok) {
+        #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = wr0 + (lane >> 4) * 4 + r;
+          if (l > 0) {
+            const float g = act_grad_from_y(prev_act, xt[row * LDSW + j]);
+            dz_nxt[row * LDSW + j] = acc[r] * g;
+          } else {
+            const int grow = row0 + row;
+            if (grow < args.batch) dx_out[(long)grow * in_d + j] = acc[r];
+          }
+        }
+      }
+    }
+    __syncthreads();
+    // swap dz ping-pong
+    const float* t = dz_cur;
+    dz_cur = dz_nxt;
+    dz_nxt = (float*)t;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // wide-net single-layer kernels: 2D grids (row blocks x 64-col groups).
 // The fused multi-layer kernel keeps every output column of a row tile
 // in one block, which leaves most of the chip idle for [*,256,256,*]
@@ -482,6 +620,13 @@ __global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
 
 // host-side dispatch over the (ROWS, MAXW) instantiations — called from
 // bindings.hip so template symbols stay in this translation unit
+void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
+                          const float* dy, float* dx, float* ws,
+                          size_t lds_bytes, int n_blocks, hipStream_t stream) {
+  hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32>), dim3(n_blocks), dim3(256),
+                     lds_bytes, stream, args, x, dy, dx, ws);
+}
+
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
                                float* out, int batch, int in_d, int out_d,
                                int act, hipStream_t stream) {
