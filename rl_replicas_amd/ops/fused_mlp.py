@@ -41,7 +41,7 @@ _ACT_CODE = {
     nn.ReLU: ACT_RELU,
 }
 
-MAX_WIDTH = 1024  # widest layer the LDS-staged kernel supports
+MAX_WIDTH = 256  # widest layer the LDS-staged kernels support
 
 
 def _extract_layers(mlp) -> Optional[tuple]:
