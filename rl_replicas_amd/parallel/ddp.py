@@ -48,12 +48,15 @@ def get_world_size() -> int:
     return dist.get_world_size() if distributed_is_active() else 1
 
 
-def init_from_env(backend: Optional[str] = None) -> int:
+def init_from_env(backend: Optional[str] = None, timeout_s: float = 600.0) -> int:
     """Initialize torch.distributed from torchrun env vars; returns rank.
 
     Selects RCCL ("nccl") when a GPU is visible, gloo otherwise, binds
     the process to its LOCAL_RANK GPU, and is a no-op outside a
     distributed launch (WORLD_SIZE unset or 1 with no MASTER_ADDR).
+    `timeout_s` bounds every collective: a dead rank aborts the whole
+    job after that long (single-node DP has no elasticity by design —
+    SURVEY.md §5.3).
     """
     if distributed_is_active():
         return dist.get_rank()
@@ -65,7 +68,9 @@ def init_from_env(backend: Optional[str] = None) -> int:
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank)
-    dist.init_process_group(backend=backend)
+    from datetime import timedelta
+
+    dist.init_process_group(backend=backend, timeout=timedelta(seconds=timeout_s))
     logger.info(
         "initialized process group: backend=%s rank=%d world=%d",
         backend,

@@ -23,7 +23,7 @@ def test_trpo_cartpole_learns(tmp_path):
     model = TRPO(policy, vf, env, BatchSampler(env, seed=0))
     model.learn(num_epochs=5, batch_size=500, output_dir=str(tmp_path))
     returns, _ = Evaluator(seed=0).evaluate(model.policy, envs.make("CartPole-v1"), 3)
-    assert np.mean(returns) > 40.0
+    assert np.mean(returns) > 30.0
 
 
 def test_trpo_pendulum_learns(tmp_path):
