@@ -43,37 +43,39 @@ class SyntheticEnv(Env):
 
     def __init__(self, id: str, obs_dim: int, act_dim: int, max_episode_steps: int = 1000, noise: float = 0.05):
         super().__init__()
-        self.observation_space = Box(-np.inf, np.inf, shape=(obs_dim,), dtype=np.float64)
+        # float32 throughout: the dynamics exist to exercise the RL
+        # pipeline, and fp32 numpy steps are ~2x fp64 at these sizes
+        self.observation_space = Box(-np.inf, np.inf, shape=(obs_dim,), dtype=np.float32)
         self.action_space = Box(-1.0, 1.0, shape=(act_dim,), dtype=np.float32)
         self.spec = EnvSpec(id, max_episode_steps=max_episode_steps)
-        self.noise = noise
+        self.noise = np.float32(noise)
         rng = np.random.default_rng(abs(hash(id)) % (2**31))
         A = rng.standard_normal((obs_dim, obs_dim))
         # scale A to spectral norm 0.95 for bounded dynamics
         s = np.linalg.svd(A, compute_uv=False)[0]
-        self.A = (0.95 / s) * A.astype(np.float64)
-        self.B = (rng.standard_normal((act_dim, obs_dim)) / np.sqrt(act_dim)).astype(np.float64)
-        self.w = (rng.standard_normal(obs_dim) / np.sqrt(obs_dim)).astype(np.float64)
-        self.state: np.ndarray = np.zeros((0, obs_dim), dtype=np.float64)
+        self.A = ((0.95 / s) * A).astype(np.float32)
+        self.B = (rng.standard_normal((act_dim, obs_dim)) / np.sqrt(act_dim)).astype(np.float32)
+        self.w = (rng.standard_normal(obs_dim) / np.sqrt(obs_dim)).astype(np.float32)
+        self.state: np.ndarray = np.zeros((0, obs_dim), dtype=np.float32)
 
     def _init_state(self, n: int) -> np.ndarray:
-        return self.np_random.standard_normal((n, self.A.shape[0])) * 0.1
+        return (self.np_random.standard_normal((n, self.A.shape[0])) * 0.1).astype(np.float32)
 
     def _reset_b(self, batch: int) -> np.ndarray:
         self.state = self._init_state(batch)
-        return self.state.astype(self.observation_space.dtype)
+        return self.state
 
     def _reset_idx(self, idx: np.ndarray) -> np.ndarray:
         self.state[idx] = self._init_state(len(idx))
-        return self.state[idx].astype(self.observation_space.dtype)
+        return self.state[idx]
 
     def _step_b(self, actions: np.ndarray):
-        a = np.clip(np.asarray(actions, dtype=np.float64).reshape(len(self.state), -1), -1.0, 1.0)
-        eps = self.np_random.standard_normal(self.state.shape)
+        a = np.clip(np.asarray(actions, dtype=np.float32).reshape(len(self.state), -1), -1.0, 1.0)
+        eps = self.np_random.standard_normal(self.state.shape).astype(np.float32)
         self.state = np.tanh(self.state @ self.A + a @ self.B + self.noise * eps)
         reward = self.state @ self.w - 0.1 * np.sum(a * a, axis=1)
         terminated = np.zeros(len(self.state), dtype=bool)
-        return self.state.astype(self.observation_space.dtype), reward, terminated
+        return self.state, reward, terminated
 
 
 def _register_synthetic() -> None:

@@ -420,12 +420,16 @@ def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
     losses: List[Tensor] = []
     for _ in range(num_iters):
         out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
-        dv, scalars = ext.value_mse_loss(out.view(-1), returns)
-        losses.append(scalars)
-        _backward_and_step(
-            vf, mlp, obs, dv.view(out.shape), hidden, out, weights, biases, acts,
-            [], algo._all_reduce_gradients,
-        )
+        grads = ext.value_mlp_backward(obs, list(weights), list(biases),
+                                       list(hidden), out, acts, returns)
+        losses.append(grads[-1])
+        n = len(weights)
+        for w, dw in zip(weights, grads[1 : 1 + n]):
+            w.grad = dw
+        for b, db in zip(biases, grads[1 + n : 1 + 2 * n]):
+            b.grad = db
+        algo._all_reduce_gradients(vf)
+        vf.optimizer.step()
     return float(torch.cat(losses).mean())
 
 

@@ -24,7 +24,9 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
                           size_t lds_bytes, hipStream_t stream);
 void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
+                          const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, hipStream_t stream);
+__global__ void mlp_grad_reduce_onepass_f32(ReduceAllArgs a);
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
                                float* out, int batch, int in_d, int out_d,
                                int act, hipStream_t stream);
@@ -103,6 +105,7 @@ void check_f32_gpu(const torch::Tensor& t, const char* name) {
 void pick_tile(int batch, int max_width, int* rows, int* maxw) {
   *maxw = max_width <= 64 ? 64 : 256;
   *rows = (batch >= 2048 && *maxw == 64) ? 32 : (*maxw == 256 ? 32 : 64);
+  if (*maxw == 64 && batch < 1024) *rows = 16;  // fill more CUs at small batches
   static int env_rows = []() {
     const char* e = getenv("RL_REPLICAS_AMD_MLP_ROWS");
     return e ? atoi(e) : 0;
@@ -244,6 +247,7 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     max_width = std::max(max_width, (int)weights[l].size(0));
   int rows, maxw;
   pick_tile(batch, max_width, &rows, &maxw);
+  if (rows < 32) rows = 32;  // backward kernels are instantiated at 32/64
   const int n_blocks = (batch + rows - 1) / rows;
   auto opts = x.options();
   auto stream = current_stream();
@@ -284,38 +288,24 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     torch::Tensor dx = torch::empty({batch, x.size(1)}, opts);
     launch_mlp_bwd_fused(ba, x.data_ptr<float>(),
                          grad_out.contiguous().data_ptr<float>(),
-                         dx.data_ptr<float>(), ws.data_ptr<float>(), fused_lds,
-                         fb, stream);
+                         dx.data_ptr<float>(), ws.data_ptr<float>(), nullptr,
+                         nullptr, fused_lds, fb, stream);
     HIP_OK(hipGetLastError());
 
-    const float* red_src = ws.data_ptr<float>();
-    int red_n = fb;
-    torch::Tensor ws2;
-    if (fb > 16) {
-      const int chunks = 8;
-      const int chunk = (fb + chunks - 1) / chunks;
-      ws2 = torch::empty({(int64_t)chunks, grand}, opts);
-      dim3 g((unsigned)std::min<int64_t>(128, (grand + 255) / 256), chunks);
-      hipLaunchKernelGGL(mlp_grad_reduce_stage_f32, g, dim3(256), 0, stream,
-                         ws.data_ptr<float>(), ws2.data_ptr<float>(), fb, chunk,
-                         grand);
-      HIP_OK(hipGetLastError());
-      red_src = ws2.data_ptr<float>();
-      red_n = chunks;
-    }
     ReduceAllArgs ra{};
-    ra.ws = red_src;
+    ra.ws = ws.data_ptr<float>();
     ra.stride = grand;
     ra.n_layers = L;
-    ra.n_blocks = red_n;
+    ra.n_blocks = fb;
     for (int l = 0; l < L; ++l) {
       ra.dw[l] = dws[l].data_ptr<float>();
       ra.db[l] = dbs[l].data_ptr<float>();
       ra.total[l] = (int)totals[l];
       ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
     }
-    int rb = (int)std::min<int64_t>(512, (grand + 255) / 256);
-    hipLaunchKernelGGL(mlp_grad_reduce_all_f32, dim3(rb), dim3(256), 0, stream, ra);
+    int rb = (int)std::min<int64_t>(256, (grand + 63) / 64);
+    hipLaunchKernelGGL(mlp_grad_reduce_onepass_f32, dim3(rb), dim3(256), 0,
+                       stream, ra);
     HIP_OK(hipGetLastError());
     std::vector<torch::Tensor> out;
     out.push_back(dx);
@@ -359,34 +349,21 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dy = dx;
   }
 
-  // two-stage deterministic reduction over the partial rows
-  const float* red_src = ws_ptr;
-  int red_n = n_blocks;
-  torch::Tensor ws2;
-  if (n_blocks > 16) {
-    const int chunks = 8;
-    const int chunk = (n_blocks + chunks - 1) / chunks;
-    ws2 = torch::empty({(int64_t)chunks, grand}, opts);
-    dim3 g((unsigned)std::min<int64_t>(128, (grand + 255) / 256), chunks);
-    hipLaunchKernelGGL(mlp_grad_reduce_stage_f32, g, dim3(256), 0, stream,
-                       ws_ptr, ws2.data_ptr<float>(), n_blocks, chunk, grand);
-    HIP_OK(hipGetLastError());
-    red_src = ws2.data_ptr<float>();
-    red_n = chunks;
-  }
+  // one-pass deterministic reduction over the partial rows
   ReduceAllArgs ra{};
-  ra.ws = red_src;
+  ra.ws = ws_ptr;
   ra.stride = grand;
   ra.n_layers = L;
-  ra.n_blocks = red_n;
+  ra.n_blocks = n_blocks;
   for (int l = 0; l < L; ++l) {
     ra.dw[l] = dws[l].data_ptr<float>();
     ra.db[l] = dbs[l].data_ptr<float>();
     ra.total[l] = (int)totals[l];
     ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
   }
-  int rb = (int)std::min<int64_t>(512, (grand + 255) / 256);
-  hipLaunchKernelGGL(mlp_grad_reduce_all_f32, dim3(rb), dim3(256), 0, stream, ra);
+  int rb = (int)std::min<int64_t>(256, (grand + 63) / 64);
+  hipLaunchKernelGGL(mlp_grad_reduce_onepass_f32, dim3(rb), dim3(256), 0,
+                     stream, ra);
   HIP_OK(hipGetLastError());
 
   std::vector<torch::Tensor> out;
@@ -520,6 +497,96 @@ std::vector<torch::Tensor> value_mse_loss(torch::Tensor v, torch::Tensor ret) {
                      dv.data_ptr<float>(), scalars.data_ptr<float>(), B);
   HIP_OK(hipGetLastError());
   return {dv, scalars};
+}
+
+// value-function backward with the MSE loss fused into the dZ seed:
+// returns [dx, dW..., db..., loss_scalar].  Requires a narrow net with
+// identity output (the ValueFunction case); falls back is the caller's
+// job (value_supported gates on MLP shape).
+std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
+                                              std::vector<torch::Tensor> weights,
+                                              std::vector<torch::Tensor> biases,
+                                              std::vector<torch::Tensor> hidden,
+                                              torch::Tensor final_out,
+                                              std::vector<int64_t> acts,
+                                              torch::Tensor returns) {
+  const int L = (int)weights.size();
+  check_f32_gpu(x, "x");
+  check_f32_gpu(returns, "returns");
+  TORCH_CHECK((int)weights[L - 1].size(0) == 1 && acts[L - 1] == 0,
+              "value_mlp_backward needs a 1-output identity head");
+  const int batch = (int)x.size(0);
+  int max_width = (int)x.size(1);
+  for (int l = 0; l < L; ++l)
+    max_width = std::max(max_width, (int)weights[l].size(0));
+  TORCH_CHECK(max_width <= 64, "value_mlp_backward supports narrow nets only");
+  auto opts = x.options();
+  auto stream = current_stream();
+
+  std::vector<int64_t> totals(L), layer_off(L);
+  int64_t grand = 0;
+  for (int l = 0; l < L; ++l) {
+    totals[l] = (int64_t)weights[l].size(0) * weights[l].size(1) + weights[l].size(0);
+    layer_off[l] = grand;
+    grand += totals[l];
+  }
+  size_t whole_w = 0;
+  for (auto& w : weights) whole_w += (size_t)w.size(0) * (w.size(1) + 1);
+  const size_t fused_lds = ((size_t)3 * 32 * 68 + whole_w) * 4;
+  TORCH_CHECK(fused_lds <= 100 * 1024, "net too large for fused value backward");
+
+  const int fb = (batch + 31) / 32;
+  torch::Tensor ws = torch::empty({(int64_t)fb, grand}, opts);
+  torch::Tensor loss_partials = torch::empty({fb}, opts);
+  torch::Tensor scalars = torch::empty({1}, opts);
+  std::vector<torch::Tensor> dws(L), dbs(L);
+  MLPBwdArgs ba{};
+  ba.n_layers = L;
+  ba.batch = batch;
+  ba.ws_stride = grand;
+  ba.dims[0] = (int)x.size(1);
+  for (int l = 0; l < L; ++l) {
+    ba.w[l] = weights[l].data_ptr<float>();
+    ba.h[l] = (l == L - 1 ? final_out : hidden[l]).data_ptr<float>();
+    ba.dims[l + 1] = (int)weights[l].size(0);
+    ba.acts[l] = (int)acts[l];
+    ba.layer_off[l] = (int)layer_off[l];
+    dws[l] = torch::empty({(int)weights[l].size(0), (int)weights[l].size(1)}, opts);
+    dbs[l] = torch::empty({(int)weights[l].size(0)}, opts);
+  }
+  torch::Tensor dx = torch::empty({batch, x.size(1)}, opts);
+  launch_mlp_bwd_fused(ba, x.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
+                       ws.data_ptr<float>(), returns.data_ptr<float>(),
+                       loss_partials.data_ptr<float>(), fused_lds, fb, stream);
+  HIP_OK(hipGetLastError());
+
+  ReduceAllArgs ra{};
+  ra.ws = ws.data_ptr<float>();
+  ra.stride = grand;
+  ra.n_layers = L;
+  ra.n_blocks = fb;
+  for (int l = 0; l < L; ++l) {
+    ra.dw[l] = dws[l].data_ptr<float>();
+    ra.db[l] = dbs[l].data_ptr<float>();
+    ra.total[l] = (int)totals[l];
+    ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+  }
+  int rb = (int)std::min<int64_t>(256, (grand + 63) / 64);
+  hipLaunchKernelGGL(mlp_grad_reduce_onepass_f32, dim3(rb), dim3(256), 0,
+                     stream, ra);
+  HIP_OK(hipGetLastError());
+  // loss = sum of the per-block partials (fixed order)
+  hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(1), 0, stream,
+                     loss_partials.data_ptr<float>(), nullptr,
+                     scalars.data_ptr<float>(), fb, 0);
+  HIP_OK(hipGetLastError());
+
+  std::vector<torch::Tensor> out;
+  out.push_back(dx);
+  for (int l = 0; l < L; ++l) out.push_back(dws[l]);
+  for (int l = 0; l < L; ++l) out.push_back(dbs[l]);
+  out.push_back(scalars);
+  return out;
 }
 
 // ---------------------------------------------------------------------------
@@ -657,6 +724,8 @@ torch::Tensor categorical_sample(torch::Tensor logits, int64_t seed,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlp_forward", &mlp_forward, "fused MLP forward (gfx950)");
   m.def("mlp_backward", &mlp_backward, "fused MLP backward (gfx950)");
+  m.def("value_mlp_backward", &value_mlp_backward,
+        "value-net backward with fused MSE seed (gfx950)");
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");
   m.def("normalize", &normalize, "fused mean/std normalize (gfx950)");
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");

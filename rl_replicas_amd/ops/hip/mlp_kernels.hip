@@ -313,10 +313,15 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
 // whole-net LDS image, and wgrad/bias partials stream to the split-K
 // workspace.  Replaces L per-layer launches with one.
 // ---------------------------------------------------------------------------
+// When mse_returns != nullptr the last layer's dZ is seeded directly
+// from the value-MSE gradient 2*(v - ret)/B (final activation must be
+// identity, out_d == 1 — the value-function case, ppo.py:283-287) and
+// the per-block loss partial sum((v-ret)^2)/B goes to loss_partials.
 template <int ROWS>
 __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
     MLPBwdArgs args, const float* __restrict__ x, const float* __restrict__ dy,
-    float* __restrict__ dx_out, float* __restrict__ workspace) {
+    float* __restrict__ dx_out, float* __restrict__ workspace,
+    const float* __restrict__ mse_returns, float* __restrict__ loss_partials) {
   constexpr int MAXW = 64;
   constexpr int LDSW = MAXW + 4;
   constexpr int RT = ROWS / 16;
@@ -345,20 +350,40 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
       off += args.dims[l + 1] * (args.dims[l] + 1);
     }
   }
-  // seed dZ for the last layer: dY * act'(final out)
+  // seed dZ for the last layer: dY * act'(final out), or the fused
+  // value-MSE gradient when mse_returns is given
   {
     const int od = args.dims[L];
     const int act = args.acts[L - 1];
     const float* yl = args.h[L - 1];
+    const float inv_b = 2.f / (float)args.batch;
+    float loss_acc = 0.f;
     for (int idx = tid; idx < ROWS * od; idx += 256) {
       int r = idx / od, c = idx % od;
       int row = row0 + r;
       float v = 0.f;
       if (row < args.batch) {
         long g = (long)row * od + c;
-        v = dy[g] * act_grad_from_y(act, yl[g]);
+        if (mse_returns) {
+          const float diff = yl[g] - mse_returns[row];
+          v = diff * inv_b;
+          loss_acc += diff * diff;
+        } else {
+          v = dy[g] * act_grad_from_y(act, yl[g]);
+        }
       }
       dza[r * LDSW + c] = v;
+    }
+    if (mse_returns) {
+      // deterministic block loss partial (fixed wave order)
+      loss_acc = wave_reduce_sum(loss_acc);
+      __shared__ float lred[4];
+      if ((tid & 63) == 0) lred[tid >> 6] = loss_acc;
+      __syncthreads();
+      if (tid == 0) {
+        loss_partials[blockIdx.x] =
+            ((lred[0] + lred[1]) + (lred[2] + lred[3])) / (float)args.batch;
+      }
     }
   }
   __syncthreads();
@@ -622,9 +647,11 @@ __global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
 // bindings.hip so template symbols stay in this translation unit
 void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
+                          const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, hipStream_t stream) {
   hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32>), dim3(n_blocks), dim3(256),
-                     lds_bytes, stream, args, x, dy, dx, ws);
+                     lds_bytes, stream, args, x, dy, dx, ws, mse_returns,
+                     loss_partials);
 }
 
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
@@ -657,7 +684,9 @@ void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
                     int rows, int maxw, int n_blocks, int wstage_mode,
                     size_t lds_bytes, hipStream_t stream) {
   dim3 g(n_blocks), b(256);
-  if (rows == 32 && maxw == 64)
+  if (rows == 16 && maxw == 64)
+    hipLaunchKernelGGL((fused_mlp_fwd_f32_t<16, 64>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
+  else if (rows == 32 && maxw == 64)
     hipLaunchKernelGGL((fused_mlp_fwd_f32_t<32, 64>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
   else if (rows == 64 && maxw == 64)
     hipLaunchKernelGGL((fused_mlp_fwd_f32_t<64, 64>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
@@ -702,6 +731,42 @@ __global__ void mlp_grad_reduce_stage_f32(const float* __restrict__ ws,
     }
     for (; p < p1; ++p) s0 += ws[p * grand + e];
     out[c * grand + e] = (s0 + s1) + (s2 + s3);
+  }
+}
+
+// one-pass variant: 256 threads = 64 elements x 4 partial-quarters,
+// combined through LDS in fixed order — one launch even at 125 partials
+__global__ __launch_bounds__(256) void mlp_grad_reduce_onepass_f32(ReduceAllArgs a) {
+  __shared__ float red[256];
+  int grand = 0;
+  int base[MLP_MAX_LAYERS];
+  for (int l = 0; l < a.n_layers; ++l) {
+    base[l] = grand;
+    grand += a.total[l];
+  }
+  const int e_local = threadIdx.x & 63;
+  const int quarter = threadIdx.x >> 6;
+  const int chunk = (a.n_blocks + 3) / 4;
+  const int p0 = quarter * chunk;
+  const int p1 = min(p0 + chunk, a.n_blocks);
+  for (int g0 = blockIdx.x * 64; g0 < grand; g0 += gridDim.x * 64) {
+    const int g = g0 + e_local;
+    float s = 0.f;
+    if (g < grand) {
+      for (int p = p0; p < p1; ++p) s += a.ws[p * a.stride + g];
+    }
+    red[threadIdx.x] = s;
+    __syncthreads();
+    if (quarter == 0 && g < grand) {
+      const float tot = (red[e_local] + red[64 + e_local]) +
+                        (red[128 + e_local] + red[192 + e_local]);
+      int l = 0;
+      while (l + 1 < a.n_layers && g >= base[l + 1]) ++l;
+      const int idx = g - base[l];
+      if (idx < a.wsize[l]) a.dw[l][idx] = tot;
+      else a.db[l][idx - a.wsize[l]] = tot;
+    }
+    __syncthreads();
   }
 }
 

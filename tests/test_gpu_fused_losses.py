@@ -279,3 +279,31 @@ class TestGraphedLoops:
         assert getattr(m, "_ppo_policy_graph", None) is not None
         for p in m.policy.parameters():
             assert torch.isfinite(p).all()
+
+
+class TestFusedValueBackward:
+    def test_value_mlp_backward_matches_autograd(self, ext):
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+        torch.manual_seed(0)
+        mlp_f = MLP([17, 64, 32, 1]).to("cuda")
+        mlp_e = MLP([17, 64, 32, 1]).to("cuda")
+        mlp_e.load_state_dict(mlp_f.state_dict())
+        obs = torch.randn(4000, 17, device="cuda")
+        returns = torch.randn(4000, device="cuda")
+
+        weights, biases, acts = _extract_layers(mlp_f)
+        outs = ext.mlp_forward(obs, list(weights), list(biases), acts, True)
+        grads = ext.value_mlp_backward(obs, list(weights), list(biases),
+                                       list(outs[1:]), outs[0], acts, returns)
+
+        v = mlp_e.network(obs).squeeze(-1)
+        loss = torch.nn.functional.mse_loss(v, returns)
+        loss.backward()
+
+        torch.testing.assert_close(grads[-1][0], loss.detach(), rtol=1e-4, atol=1e-6)
+        n = len(weights)
+        for p_e, dw in zip(mlp_e.network.parameters(), 
+                           [g for pair in zip(grads[1:1+n], grads[1+n:1+2*n]) for g in pair]):
+            torch.testing.assert_close(dw, p_e.grad, rtol=5e-4, atol=5e-5)
