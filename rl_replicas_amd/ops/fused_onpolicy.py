@@ -119,11 +119,7 @@ def _old_logp(policy, kind: str, obs: Tensor, actions: Tensor) -> Tensor:
 def _graphs_common(algo) -> bool:
     import os
 
-    if os.environ.get("RL_REPLICAS_AMD_DISABLE_GRAPHS", "0") == "1":
-        return False
-    from rl_replicas_amd.parallel.ddp import distributed_is_active
-
-    return not (distributed_is_active() or getattr(algo, "_dp_enabled", False))
+    return os.environ.get("RL_REPLICAS_AMD_DISABLE_GRAPHS", "0") != "1"
 
 
 def _graphs_enabled(algo) -> bool:
@@ -136,6 +132,18 @@ def _graphs_enabled_value(algo) -> bool:
     from rl_replicas_amd.ops.fused_adam import FusedAdam
 
     return _graphs_common(algo) and isinstance(algo.value_function.optimizer, FusedAdam)
+
+
+def _dp_active(algo) -> bool:
+    import os
+
+    from rl_replicas_amd.parallel.ddp import distributed_is_active
+
+    return (
+        distributed_is_active()
+        or getattr(algo, "_dp_enabled", False)
+        or os.environ.get("RL_REPLICAS_AMD_FORCE_DP_GRAPHS", "0") == "1"
+    )
 
 
 def _ensure_adam_state(optimizer) -> List[Tensor]:
@@ -180,14 +188,20 @@ class _CapturedLoop:
 
 class _GraphedPPO:
     """One captured PPO policy iteration: fwd + loss + bwd + Adam +
-    KL-eval forward, replayed up to num_policy_gradients times."""
+    KL-eval forward, replayed up to num_policy_gradients times.
+
+    `split=True` (data parallelism) captures TWO graphs — grads before
+    / Adam+KL after — and runs the eager gradient all-reduce between
+    them on the capture-stable .grad buffers; collectives are never
+    captured."""
 
     def __init__(self, algo, kind: str, obs0: Tensor, actions0: Tensor,
-                 adv0: Tensor, old_logp0: Tensor):
+                 adv0: Tensor, old_logp0: Tensor, split: bool):
         ext = ops._load_extension()
         policy = algo.policy
         mlp = _mlp_of(policy)
         self.kind = kind
+        self.split = split
         # capture-stable input buffers, pre-filled with real data so the
         # warmup runs on valid values
         self.obs = obs0.clone()
@@ -196,7 +210,7 @@ class _GraphedPPO:
         self.old_logp = old_logp0.clone()
         clip = float(algo.clip_range)
 
-        def body():
+        def body_pre():
             out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
             if kind == "gaussian":
                 dmean, dlog_std, scalars = ext.gaussian_policy_loss(
@@ -211,33 +225,60 @@ class _GraphedPPO:
                 )
                 extra = []
                 grad_out = dlogits
-            _backward_and_step(
-                policy, mlp, self.obs, grad_out, hidden, out, weights, biases,
-                acts, extra, lambda _m: None,
-            )
+            grads = ext.mlp_backward(grad_out, self.obs, list(weights),
+                                     list(biases), list(hidden), out, acts)
+            n = len(weights)
+            for w, dw in zip(weights, grads[1 : 1 + n]):
+                w.grad = dw
+            for b, db in zip(biases, grads[1 + n :]):
+                b.grad = db
+            for param, grad in extra:
+                param.grad = grad
+            return scalars
+
+        def body_post():
+            policy.optimizer.step()
             new_out = _forward_only(mlp, self.obs)
             if kind == "gaussian":
-                kl = ext.gaussian_kl(new_out, self.actions, policy.log_std.data, self.old_logp)
+                kl = ext.gaussian_kl(new_out, self.actions, policy.log_std.data,
+                                     self.old_logp)
             else:
                 kl = ext.categorical_kl(new_out, self.actions, self.old_logp)
-            return scalars, kl
+            return kl
 
         state = [p.data for p in policy.parameters()]
         state += _ensure_adam_state(policy.optimizer)
-        self.loop = _CapturedLoop(body, state)
+        if split:
+            # order matters: pre's capture pins the .grad buffers that
+            # post's Adam capture reads
+            self.pre = _CapturedLoop(body_pre, state)
+            self.post = _CapturedLoop(body_post, state)
+        else:
+            def body():
+                scalars = body_pre()
+                kl = body_post()
+                return scalars, kl
+
+            self.loop = _CapturedLoop(body, state)
 
     def run(self, algo, obs, actions, advantages, old_logp) -> Dict[str, float]:
         self.obs.copy_(obs)
         self.actions.copy_(actions.view(self.actions.shape))
         self.adv.copy_(advantages)
         self.old_logp.copy_(old_logp)
+        policy = algo.policy
         loss_before = None
         approximate_kl = 0.0
         for i in range(algo.num_policy_gradients):
-            scalars, kl = self.loop.replay()
+            if self.split:
+                scalars = self.pre.replay()
+                algo._all_reduce_gradients(policy)
+                kl = self.post.replay()
+            else:
+                scalars, kl = self.loop.replay()
             if loss_before is None:
                 loss_before = float(scalars[0])
-            approximate_kl = float(kl[0])
+            approximate_kl = float(algo._reduce_scalar_mean(kl[0]))
             if approximate_kl > 1.5 * algo.max_kl_divergence:
                 logger.info(
                     "Early stopping at update %d due to reaching max KL divergence.", i
@@ -247,35 +288,56 @@ class _GraphedPPO:
 
 
 class _GraphedValueLoop:
-    """The whole num_value_gradients value-function loop as ONE graph."""
+    """The whole num_value_gradients value-function loop as ONE graph
+    (single-process), or per-iteration pre/post graphs around the
+    eager gradient all-reduce (data parallelism)."""
 
-    def __init__(self, algo, obs0: Tensor, returns0: Tensor, num_iters: int):
+    def __init__(self, algo, obs0: Tensor, returns0: Tensor, num_iters: int,
+                 split: bool):
         ext = ops._load_extension()
         vf = algo.value_function
         mlp = vf.network
         self.obs = obs0.clone()
         self.returns = returns0.clone()
         self.num_iters = num_iters
+        self.split = split
 
-        def body():
-            losses = []
-            for _ in range(num_iters):
-                out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
-                dv, scalars = ext.value_mse_loss(out.view(-1), self.returns)
-                losses.append(scalars)
-                _backward_and_step(
-                    vf, mlp, self.obs, dv.view(out.shape), hidden, out, weights,
-                    biases, acts, [], lambda _m: None,
-                )
-            return losses
+        def iter_pre():
+            out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
+            grads = ext.value_mlp_backward(self.obs, list(weights), list(biases),
+                                           list(hidden), out, acts, self.returns)
+            n = len(weights)
+            for w, dw in zip(weights, grads[1 : 1 + n]):
+                w.grad = dw
+            for b, db in zip(biases, grads[1 + n : 1 + 2 * n]):
+                b.grad = db
+            return grads[-1]
 
         state = [p.data for p in vf.parameters()]
         state += _ensure_adam_state(vf.optimizer)
-        self.loop = _CapturedLoop(body, state)
+        if split:
+            self.pre = _CapturedLoop(iter_pre, state)
+            self.post = _CapturedLoop(lambda: vf.optimizer.step(), state)
+        else:
+            def body():
+                losses = []
+                for _ in range(num_iters):
+                    losses.append(iter_pre())
+                    vf.optimizer.step()
+                return losses
 
-    def run(self, obs, returns) -> float:
+            self.loop = _CapturedLoop(body, state)
+
+    def run(self, algo, obs, returns) -> float:
         self.obs.copy_(obs)
         self.returns.copy_(returns)
+        if self.split:
+            losses = []
+            for _ in range(self.num_iters):
+                losses.append(self.pre.replay())
+                algo._all_reduce_gradients(algo.value_function)
+                self.post.replay()
+            return float(torch.cat(losses).mean())
         losses = self.loop.replay()
         return float(torch.cat(losses).mean())
 
@@ -309,11 +371,12 @@ def ppo_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[s
         old_logp = _old_logp(algo.old_policy, kind, obs, actions_k)
 
     if _graphs_enabled(algo):
-        key = (kind, tuple(obs.shape), tuple(actions_k.shape),
+        split = _dp_active(algo)
+        key = (kind, tuple(obs.shape), tuple(actions_k.shape), split,
                tuple(id(p) for p in policy.parameters()))
         graphed = _get_cached_graph(
             algo, "_ppo_policy_graph", key,
-            lambda: _GraphedPPO(algo, kind, obs, actions_k, advantages, old_logp),
+            lambda: _GraphedPPO(algo, kind, obs, actions_k, advantages, old_logp, split),
         )
         metrics = graphed.run(algo, obs, actions_k, advantages, old_logp)
         algo.old_policy.load_state_dict(policy.state_dict())
@@ -411,12 +474,14 @@ def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
     from rl_replicas_amd.ops.fused_adam import FusedAdam
 
     if _graphs_enabled_value(algo):
-        key = (tuple(obs.shape), num_iters, tuple(id(p) for p in vf.parameters()))
+        split = _dp_active(algo)
+        key = (tuple(obs.shape), num_iters, split,
+               tuple(id(p) for p in vf.parameters()))
         graphed = _get_cached_graph(
             algo, "_value_graph", key,
-            lambda: _GraphedValueLoop(algo, obs, returns, num_iters),
+            lambda: _GraphedValueLoop(algo, obs, returns, num_iters, split),
         )
-        return graphed.run(obs, returns)
+        return graphed.run(algo, obs, returns)
     losses: List[Tensor] = []
     for _ in range(num_iters):
         out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
