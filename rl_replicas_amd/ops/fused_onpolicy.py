@@ -334,7 +334,9 @@ class _GraphedValueLoop:
         if self.split:
             losses = []
             for _ in range(self.num_iters):
-                losses.append(self.pre.replay())
+                # clone: the captured output buffer is overwritten by the
+                # next replay
+                losses.append(self.pre.replay().clone())
                 algo._all_reduce_gradients(algo.value_function)
                 self.post.replay()
             return float(torch.cat(losses).mean())
