@@ -64,7 +64,15 @@ def main() -> None:
     parser.add_argument("--batch-per-gpu", type=int, default=4000)
     parser.add_argument("--num-envs", type=int, default=200)
     parser.add_argument("--phase-timing", action="store_true", help="print sample/train ms split (rank 0, stderr)")
+    parser.add_argument(
+        "--dtype",
+        choices=["fp32", "bf16"],
+        default="fp32",
+        help="MLP GEMM compute dtype: exact-fp32 MFMA (default, exceeds the "
+        "reference's fp32) or bf16 MFMA with fp32 accumulate",
+    )
     args = parser.parse_args()
+    os.environ["RL_REPLICAS_AMD_COMPUTE_DTYPE"] = args.dtype
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -151,7 +159,7 @@ def main() -> None:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": value / 946.0,
-            "dtype": "fp32",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": "PPO HalfCheetah-v4 (policy MLP [17,64,32,6] tanh, value [17,64,32,1])",

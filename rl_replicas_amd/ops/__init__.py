@@ -53,6 +53,20 @@ def _allow_eager() -> bool:
     return os.environ.get("RL_REPLICAS_AMD_ALLOW_EAGER", "0") == "1"
 
 
+def compute_bf16() -> int:
+    """1 when the MLP GEMM compute dtype is bf16 (MFMA
+    v_mfma_f32_16x16x32_bf16, fp32 accumulate) instead of exact fp32
+    MFMA.  Set RL_REPLICAS_AMD_COMPUTE_DTYPE=bf16 (or call
+    set_compute_dtype) before building models; loss/scan/optimizer
+    math stays fp32 (mixed precision)."""
+    return 1 if os.environ.get("RL_REPLICAS_AMD_COMPUTE_DTYPE", "fp32") == "bf16" else 0
+
+
+def set_compute_dtype(dtype: str) -> None:
+    assert dtype in ("fp32", "bf16"), dtype
+    os.environ["RL_REPLICAS_AMD_COMPUTE_DTYPE"] = dtype
+
+
 def wants_hip(t) -> bool:
     """True if `t` (tensor/parameter) is on GPU and the HIP path should run.
 

@@ -108,7 +108,9 @@ def try_fused_forward(mlp, input: Tensor):
         out = _FusedMLPTrainFunction.apply(x.contiguous(), acts, *weights, *biases)
     else:
         ext = ops._load_extension()
-        out = ext.mlp_forward(x.contiguous(), list(weights), list(biases), acts, False)[0]
+        out = ext.mlp_forward(
+            x.contiguous(), list(weights), list(biases), acts, False, ops.compute_bf16()
+        )[0]
     return out.squeeze(0) if squeeze else out
 
 
@@ -127,7 +129,7 @@ class _FusedMLPTrainFunction(torch.autograd.Function):
         n = len(params) // 2
         weights = list(params[:n])
         biases = list(params[n:])
-        outs = ext.mlp_forward(x, weights, biases, acts, True)
+        outs = ext.mlp_forward(x, weights, biases, acts, True, ops.compute_bf16())
         # outs = [final_out, act_out_0, ..., act_out_{n-2}]
         ctx.save_for_backward(x, *params, *outs[1:], outs[0])
         ctx.acts = acts
@@ -147,7 +149,8 @@ class _FusedMLPTrainFunction(torch.autograd.Function):
         hidden = list(saved[1 + 2 * n : n * 3])  # n-1 hidden activations
         final_out = saved[-1]
         grads = ext.mlp_backward(
-            grad_out.contiguous(), x, weights, biases, hidden, final_out, ctx.acts
+            grad_out.contiguous(), x, weights, biases, hidden, final_out,
+            ctx.acts, ops.compute_bf16(),
         )
         # grads = [dx, dW0.., dWn-1, db0.., dbn-1]
         dx = grads[0]

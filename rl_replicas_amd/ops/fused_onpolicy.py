@@ -67,21 +67,23 @@ def _forward_saved(mlp, obs: Tensor):
     """Fused forward that saves hidden activations for backward."""
     ext = ops._load_extension()
     weights, biases, acts = _extract_layers(mlp)
-    outs = ext.mlp_forward(obs, list(weights), list(biases), acts, True)
+    outs = ext.mlp_forward(obs, list(weights), list(biases), acts, True,
+                           ops.compute_bf16())
     return outs[0], outs[1:], weights, biases, acts
 
 
 def _forward_only(mlp, obs: Tensor) -> Tensor:
     ext = ops._load_extension()
     weights, biases, acts = _extract_layers(mlp)
-    return ext.mlp_forward(obs, list(weights), list(biases), acts, False)[0]
+    return ext.mlp_forward(obs, list(weights), list(biases), acts, False,
+                           ops.compute_bf16())[0]
 
 
 def _backward_and_step(policy, mlp, obs, grad_out, hidden, final_out, weights,
                        biases, acts, extra_grads, all_reduce_hook) -> None:
     ext = ops._load_extension()
     grads = ext.mlp_backward(grad_out, obs, list(weights), list(biases),
-                             list(hidden), final_out, acts)
+                             list(hidden), final_out, acts, ops.compute_bf16())
     n = len(weights)
     for w, dw in zip(weights, grads[1 : 1 + n]):
         w.grad = dw
@@ -226,7 +228,8 @@ class _GraphedPPO:
                 extra = []
                 grad_out = dlogits
             grads = ext.mlp_backward(grad_out, self.obs, list(weights),
-                                     list(biases), list(hidden), out, acts)
+                                     list(biases), list(hidden), out, acts,
+                                     ops.compute_bf16())
             n = len(weights)
             for w, dw in zip(weights, grads[1 : 1 + n]):
                 w.grad = dw
@@ -305,7 +308,8 @@ class _GraphedValueLoop:
         def iter_pre():
             out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
             grads = ext.value_mlp_backward(self.obs, list(weights), list(biases),
-                                           list(hidden), out, acts, self.returns)
+                                           list(hidden), out, acts, self.returns,
+                                           ops.compute_bf16())
             n = len(weights)
             for w, dw in zip(weights, grads[1 : 1 + n]):
                 w.grad = dw
@@ -488,7 +492,8 @@ def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
     for _ in range(num_iters):
         out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
         grads = ext.value_mlp_backward(obs, list(weights), list(biases),
-                                       list(hidden), out, acts, returns)
+                                       list(hidden), out, acts, returns,
+                                       ops.compute_bf16())
         losses.append(grads[-1])
         n = len(weights)
         for w, dw in zip(weights, grads[1 : 1 + n]):

@@ -16,7 +16,7 @@
 // kernel declarations (defined in the .hip translation units)
 void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
                     int rows, int maxw, int n_blocks, int wstage_mode,
-                    size_t lds_bytes, hipStream_t stream);
+                    size_t lds_bytes, int compute_bf16, hipStream_t stream);
 void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
                           const float* W, float* dx, float* ws, long ws_stride,
                           int batch, int out_d, int in_d, int act, int rows,
@@ -25,7 +25,8 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
 void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
-                          size_t lds_bytes, int n_blocks, hipStream_t stream);
+                          size_t lds_bytes, int n_blocks, int compute_bf16,
+                          hipStream_t stream);
 __global__ void mlp_grad_reduce_onepass_f32(ReduceAllArgs a);
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
                                float* out, int batch, int in_d, int out_d,
@@ -162,7 +163,8 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
                                        std::vector<torch::Tensor> weights,
                                        std::vector<torch::Tensor> biases,
                                        std::vector<int64_t> acts,
-                                       bool save_hidden) {
+                                       bool save_hidden,
+                                       int64_t compute_bf16) {
   const int L = (int)weights.size();
   TORCH_CHECK(L >= 1 && L <= MLP_MAX_LAYERS, "unsupported layer count ", L);
   check_f32_gpu(x, "x");
@@ -225,7 +227,7 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
       size_t lds_bytes;
       pick_wstage(weights, rows, maxw, false, 0, &wmode, &lds_bytes);
       launch_mlp_fwd(args, x.data_ptr<float>(), save_hidden ? 1 : 0, rows, maxw,
-                     n_blocks, wmode, lds_bytes, stream);
+                     n_blocks, wmode, lds_bytes, (int)compute_bf16, stream);
       HIP_OK(hipGetLastError());
     }
   }
@@ -237,7 +239,8 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
                                         std::vector<torch::Tensor> biases,
                                         std::vector<torch::Tensor> hidden,
                                         torch::Tensor final_out,
-                                        std::vector<int64_t> acts) {
+                                        std::vector<int64_t> acts,
+                                        int64_t compute_bf16) {
   const int L = (int)weights.size();
   check_f32_gpu(grad_out, "grad_out");
   check_f32_gpu(x, "x");
@@ -289,7 +292,7 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     launch_mlp_bwd_fused(ba, x.data_ptr<float>(),
                          grad_out.contiguous().data_ptr<float>(),
                          dx.data_ptr<float>(), ws.data_ptr<float>(), nullptr,
-                         nullptr, fused_lds, fb, stream);
+                         nullptr, fused_lds, fb, (int)compute_bf16, stream);
     HIP_OK(hipGetLastError());
 
     ReduceAllArgs ra{};
@@ -509,7 +512,8 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
                                               std::vector<torch::Tensor> hidden,
                                               torch::Tensor final_out,
                                               std::vector<int64_t> acts,
-                                              torch::Tensor returns) {
+                                              torch::Tensor returns,
+                                              int64_t compute_bf16) {
   const int L = (int)weights.size();
   check_f32_gpu(x, "x");
   check_f32_gpu(returns, "returns");
@@ -557,7 +561,8 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   torch::Tensor dx = torch::empty({batch, x.size(1)}, opts);
   launch_mlp_bwd_fused(ba, x.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
                        ws.data_ptr<float>(), returns.data_ptr<float>(),
-                       loss_partials.data_ptr<float>(), fused_lds, fb, stream);
+                       loss_partials.data_ptr<float>(), fused_lds, fb,
+                       (int)compute_bf16, stream);
   HIP_OK(hipGetLastError());
 
   ReduceAllArgs ra{};
@@ -722,10 +727,18 @@ torch::Tensor categorical_sample(torch::Tensor logits, int64_t seed,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("mlp_forward", &mlp_forward, "fused MLP forward (gfx950)");
-  m.def("mlp_backward", &mlp_backward, "fused MLP backward (gfx950)");
+  m.def("mlp_forward", &mlp_forward, "fused MLP forward (gfx950)",
+        py::arg("x"), py::arg("weights"), py::arg("biases"), py::arg("acts"),
+        py::arg("save_hidden"), py::arg("compute_bf16") = 0);
+  m.def("mlp_backward", &mlp_backward, "fused MLP backward (gfx950)",
+        py::arg("grad_out"), py::arg("x"), py::arg("weights"), py::arg("biases"),
+        py::arg("hidden"), py::arg("final_out"), py::arg("acts"),
+        py::arg("compute_bf16") = 0);
   m.def("value_mlp_backward", &value_mlp_backward,
-        "value-net backward with fused MSE seed (gfx950)");
+        "value-net backward with fused MSE seed (gfx950)", py::arg("x"),
+        py::arg("weights"), py::arg("biases"), py::arg("hidden"),
+        py::arg("final_out"), py::arg("acts"), py::arg("returns"),
+        py::arg("compute_bf16") = 0);
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");
   m.def("normalize", &normalize, "fused mean/std normalize (gfx950)");
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");

@@ -8,6 +8,12 @@
 
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef __hip_bfloat16 bf16_t;
+// bf16 MFMA fragment: 8 elements per lane (v_mfma_f32_16x16x32_bf16)
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+
+DEV_INLINE __bf16 f32_to_bf16(float v) {
+  return (__bf16)v;  // hardware RNE conversion on gfx950
+}
 
 // activation codes shared with python (ops/fused_mlp.py)
 #define ACT_IDENTITY 0

@@ -72,7 +72,7 @@ DEV_INLINE void wave_lds_fence() {
 // ---------------------------------------------------------------------------
 // fused forward
 // ---------------------------------------------------------------------------
-template <int ROWS, int MAXW>
+template <int ROWS, int MAXW, bool BF16 = false>
 __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
     MLPArgs args, const float* __restrict__ x, int save_hidden, int wstage_mode) {
   constexpr int LDSW = MAXW + 4;
@@ -122,11 +122,26 @@ __global__ __launch_bounds__(256) void fused_mlp_fwd_f32_t(
       const bool jok = j < out_d;
       if (wstage_mode == 0) {
         const float* wl = wlds + woff;
-        for (int k0 = 0; k0 < in_d; k0 += 4) {
-          const int kk = k0 + k;
-          float a = (kk < in_d) ? buf_in[(wr0 + i) * LDSW + kk] : 0.f;
-          float bv = (jok && kk < in_d) ? wl[j * wrow + kk] : 0.f;
-          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        if constexpr (BF16) {
+          // bf16 compute: v_mfma_f32_16x16x32_bf16, fp32 accumulate;
+          // fragments built from the fp32 LDS images with RNE converts
+          for (int k0 = 0; k0 < in_d; k0 += 32) {
+            bf16x8 af, bf;
+            #pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              const int kk = k0 + k * 8 + e;
+              af[e] = f32_to_bf16((kk < in_d) ? buf_in[(wr0 + i) * LDSW + kk] : 0.f);
+              bf[e] = f32_to_bf16((jok && kk < in_d) ? wl[j * wrow + kk] : 0.f);
+            }
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+          }
+        } else {
+          for (int k0 = 0; k0 < in_d; k0 += 4) {
+            const int kk = k0 + k;
+            float a = (kk < in_d) ? buf_in[(wr0 + i) * LDSW + kk] : 0.f;
+            float bv = (jok && kk < in_d) ? wl[j * wrow + kk] : 0.f;
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+          }
         }
       } else if (wstage_mode == 2) {
         // direct global W reads (wide nets at large grids: the L2
@@ -317,7 +332,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
 // from the value-MSE gradient 2*(v - ret)/B (final activation must be
 // identity, out_d == 1 — the value-function case, ppo.py:283-287) and
 // the per-block loss partial sum((v-ret)^2)/B goes to loss_partials.
-template <int ROWS>
+template <int ROWS, bool BF16 = false>
 __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
     MLPBwdArgs args, const float* __restrict__ x, const float* __restrict__ dy,
     float* __restrict__ dx_out, float* __restrict__ workspace,
@@ -410,10 +425,23 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
       const int ii = it * 16 + i;
       for (int jt = 0; jt < in_d; jt += 16) {
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-        for (int k0 = 0; k0 < ROWS; k0 += 4) {
-          float a = (ii < out_d) ? dz_cur[(k0 + k) * LDSW + ii] : 0.f;
-          float bv = (jt + i < in_d) ? xt[(k0 + k) * LDSW + jt + i] : 0.f;
-          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        if constexpr (BF16) {
+          for (int k0 = 0; k0 < ROWS; k0 += 32) {
+            bf16x8 af, bf;
+            #pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              const int kk = k0 + k * 8 + e;
+              af[e] = f32_to_bf16((kk < ROWS && ii < out_d) ? dz_cur[kk * LDSW + ii] : 0.f);
+              bf[e] = f32_to_bf16((kk < ROWS && jt + i < in_d) ? xt[kk * LDSW + jt + i] : 0.f);
+            }
+            acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+          }
+        } else {
+          for (int k0 = 0; k0 < ROWS; k0 += 4) {
+            float a = (ii < out_d) ? dz_cur[(k0 + k) * LDSW + ii] : 0.f;
+            float bv = (jt + i < in_d) ? xt[(k0 + k) * LDSW + jt + i] : 0.f;
+            acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+          }
         }
         const int col = jt + i;
         if (col < in_d) {
@@ -440,11 +468,24 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
       const int j = jt + i;
       const bool jok = j < in_d;
-      for (int k0 = 0; k0 < out_d; k0 += 4) {
-        const int kk = k0 + k;
-        float a = (kk < out_d) ? dz_cur[(wr0 + i) * LDSW + kk] : 0.f;
-        float bv = (jok && kk < out_d) ? wl[kk * wrow + j] : 0.f;
-        acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+      if constexpr (BF16) {
+        for (int k0 = 0; k0 < out_d; k0 += 32) {
+          bf16x8 af, bf;
+          #pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int kk = k0 + k * 8 + e;
+            af[e] = f32_to_bf16((kk < out_d) ? dz_cur[(wr0 + i) * LDSW + kk] : 0.f);
+            bf[e] = f32_to_bf16((jok && kk < out_d) ? wl[kk * wrow + j] : 0.f);
+          }
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+        }
+      } else {
+        for (int k0 = 0; k0 < out_d; k0 += 4) {
+          const int kk = k0 + k;
+          float a = (kk < out_d) ? dz_cur[(wr0 + i) * LDSW + kk] : 0.f;
+          float bv = (jok && kk < out_d) ? wl[kk * wrow + j] : 0.f;
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
       }
       if (jok) {
         #pragma unroll
@@ -648,10 +689,16 @@ __global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
 void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
-                          size_t lds_bytes, int n_blocks, hipStream_t stream) {
-  hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32>), dim3(n_blocks), dim3(256),
-                     lds_bytes, stream, args, x, dy, dx, ws, mse_returns,
-                     loss_partials);
+                          size_t lds_bytes, int n_blocks, int compute_bf16,
+                          hipStream_t stream) {
+  if (compute_bf16)
+    hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32, true>), dim3(n_blocks),
+                       dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
+                       mse_returns, loss_partials);
+  else
+    hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32, false>), dim3(n_blocks),
+                       dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
+                       mse_returns, loss_partials);
 }
 
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
@@ -682,8 +729,17 @@ void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
 
 void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
                     int rows, int maxw, int n_blocks, int wstage_mode,
-                    size_t lds_bytes, hipStream_t stream) {
+                    size_t lds_bytes, int compute_bf16, hipStream_t stream) {
   dim3 g(n_blocks), b(256);
+  if (compute_bf16 && maxw == 64 && wstage_mode == 0) {
+    if (rows == 16)
+      hipLaunchKernelGGL((fused_mlp_fwd_f32_t<16, 64, true>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
+    else if (rows == 32)
+      hipLaunchKernelGGL((fused_mlp_fwd_f32_t<32, 64, true>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
+    else
+      hipLaunchKernelGGL((fused_mlp_fwd_f32_t<64, 64, true>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
+    return;
+  }
   if (rows == 16 && maxw == 64)
     hipLaunchKernelGGL((fused_mlp_fwd_f32_t<16, 64>), g, b, lds_bytes, stream, args, x, save_hidden, wstage_mode);
   else if (rows == 32 && maxw == 64)

@@ -255,3 +255,62 @@ class TestSampleKernels:
         torch.testing.assert_close(
             freqs, torch.tensor([0.1, 0.2, 0.7], device="cuda"), atol=0.01, rtol=0.0
         )
+
+
+class TestBF16Compute:
+    """bf16 MFMA compute path (v_mfma_f32_16x16x32_bf16, fp32 accum)
+    vs the exact-fp32 MFMA path at bf16 tolerance — also a layout check:
+    random (asymmetric) weights catch any fragment transpose."""
+
+    def test_forward_bf16_close_to_fp32(self, ext):
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+        torch.manual_seed(0)
+        mlp = MLP([17, 64, 32, 6]).to("cuda")
+        weights, biases, acts = _extract_layers(mlp)
+        x = torch.randn(4000, 17, device="cuda")
+        out32 = ext.mlp_forward(x, list(weights), list(biases), acts, False, 0)[0]
+        out16 = ext.mlp_forward(x, list(weights), list(biases), acts, False, 1)[0]
+        assert not torch.equal(out32, out16)  # bf16 path actually ran
+        torch.testing.assert_close(out16, out32, rtol=0.03, atol=0.03)
+
+    def test_backward_bf16_close_to_fp32(self, ext):
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+        torch.manual_seed(1)
+        mlp = MLP([17, 64, 32, 6]).to("cuda")
+        weights, biases, acts = _extract_layers(mlp)
+        x = torch.randn(4000, 17, device="cuda")
+        outs = ext.mlp_forward(x, list(weights), list(biases), acts, True, 0)
+        dy = torch.randn_like(outs[0])
+        g32 = ext.mlp_backward(dy, x, list(weights), list(biases), list(outs[1:]), outs[0], acts, 0)
+        g16 = ext.mlp_backward(dy, x, list(weights), list(biases), list(outs[1:]), outs[0], acts, 1)
+        for a, b in zip(g16[1:], g32[1:]):
+            torch.testing.assert_close(a, b, rtol=0.05, atol=0.02)
+
+    def test_ppo_trains_in_bf16(self, ext, monkeypatch, tmp_path):
+        import numpy as np
+
+        from rl_replicas_amd import envs, ops
+        from rl_replicas_amd.algorithms import PPO
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import GaussianPolicy
+        from rl_replicas_amd.samplers import VectorSampler
+        from rl_replicas_amd.value_function import ValueFunction
+
+        monkeypatch.setenv("RL_REPLICAS_AMD_COMPUTE_DTYPE", "bf16")
+        torch.manual_seed(0)
+        venv = envs.VectorEnv("HalfCheetah-v4", num_envs=10)
+        pnet = MLP([17, 64, 32, 6]).to("cuda")
+        log_std = nn.Parameter(-0.5 * torch.ones(6, device="cuda"))
+        policy = GaussianPolicy(
+            pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+        )
+        vnet = MLP([17, 64, 32, 1]).to("cuda")
+        vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+        model = PPO(policy, vf, venv, VectorSampler(venv, seed=0))
+        model.learn(num_epochs=3, batch_size=500, output_dir=str(tmp_path))
+        for p in pnet.parameters():
+            assert torch.isfinite(p).all()
