@@ -17,7 +17,7 @@ launch latency, which HIP-graph capture of the CG body can amortize).
 from __future__ import annotations
 
 import logging
-from typing import Callable, Iterable, List, Tuple
+from typing import Callable, Iterable, List, Optional, Tuple
 
 import torch
 from torch import Tensor
@@ -70,19 +70,31 @@ class ConjugateGradientOptimizer(Optimizer):
         self.hvp_damping_coefficient = hvp_damping_coefficient
 
     # ------------------------------------------------------------------
-    def step(self, loss_function: Callable, kl_divergence_function: Callable) -> None:  # type: ignore[override]
+    def step(
+        self,
+        loss_function: Callable,
+        kl_divergence_function: Callable,
+        fisher_vector_product: Optional[Callable] = None,
+    ) -> None:  # type: ignore[override]
         """One trust-region update.
 
         Caller must have populated `.grad` of the params with the loss
         gradient (reference trpo.py:236-240 does `loss.backward()` then
         `optimizer.step(loss_fn, kl_fn)`).
+
+        `fisher_vector_product` (extension): an analytic v -> (H + damping)v
+        callable replacing the double-backward FVP — TRPO supplies the
+        Gauss-Newton form, exact at its evaluation point where
+        policy == old_policy (algorithms/trpo.py).
         """
         params: List[Tensor] = [
             p for group in self.param_groups for p in group["params"] if p.grad is not None
         ]
         loss_grad = _flatten([p.grad for p in params]).detach()
 
-        hvp = self._make_fisher_vector_product(kl_divergence_function, params)
+        hvp = fisher_vector_product or self._make_fisher_vector_product(
+            kl_divergence_function, params
+        )
         direction = self._conjugate_gradient(hvp, loss_grad)
         # NaN guard on the direction (reference :83)
         direction = torch.nan_to_num(direction, nan=0.0)
@@ -158,7 +170,11 @@ class ConjugateGradientOptimizer(Optimizer):
         (reference :204-250).
         """
         saved = [p.detach().clone() for p in params]
-        loss_before = loss_function()
+        # line-search evaluations only need VALUES (the reference builds
+        # throwaway graphs here, CGO:204-250) — no_grad lets the fused
+        # inference kernels run
+        with torch.no_grad():
+            loss_before = loss_function()
 
         # per-parameter views of the flat step
         numels = [p.numel() for p in params]
@@ -174,8 +190,8 @@ class ConjugateGradientOptimizer(Optimizer):
             with torch.no_grad():
                 for p, p0, s in zip(params, saved, step_views):
                     p.data.copy_(p0 - ratio * s)
-            new_loss = loss_function()
-            constraint = constraint_function()
+                new_loss = loss_function()
+                constraint = constraint_function()
             if new_loss < loss_before and constraint <= self.max_constraint:
                 break
 
