@@ -80,6 +80,7 @@ __global__ void q_target_kernel(const float* r, const float* d, const float* qn,
                                 float* out, float gamma, int n);
 
 __global__ void fused_adam_kernel(AdamArgs a);
+__global__ void adam_step_bump_kernel(AdamArgs a);
 __global__ void fused_polyak_kernel(PolyakArgs a);
 
 namespace {
@@ -668,7 +669,13 @@ void fused_adam_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
     a.beta2 = (float)beta2;
     a.eps = (float)eps;
     a.weight_decay = (float)weight_decay;
-    hipLaunchKernelGGL(fused_adam_kernel, dim3(a.n_tensors), dim3(256), 0,
+    int max_chunks = 1;
+    for (int i = 0; i < a.n_tensors; ++i)
+      max_chunks = std::max(max_chunks, (a.numel[i] + 8191) / 8192);
+    hipLaunchKernelGGL(fused_adam_kernel, dim3(a.n_tensors, max_chunks),
+                       dim3(256), 0, current_stream(), a);
+    HIP_OK(hipGetLastError());
+    hipLaunchKernelGGL(adam_step_bump_kernel, dim3(1), dim3(MT_MAX_TENSORS), 0,
                        current_stream(), a);
     HIP_OK(hipGetLastError());
   }
@@ -687,8 +694,11 @@ void fused_polyak_(std::vector<torch::Tensor> srcs, std::vector<torch::Tensor> d
       a.numel[i] = (int)srcs[t].numel();
     }
     a.rho = (float)rho;
-    hipLaunchKernelGGL(fused_polyak_kernel, dim3(a.n_tensors), dim3(256), 0,
-                       current_stream(), a);
+    int max_chunks = 1;
+    for (int i = 0; i < a.n_tensors; ++i)
+      max_chunks = std::max(max_chunks, (a.numel[i] + 8191) / 8192);
+    hipLaunchKernelGGL(fused_polyak_kernel, dim3(a.n_tensors, max_chunks),
+                       dim3(256), 0, current_stream(), a);
     HIP_OK(hipGetLastError());
   }
 }

@@ -540,20 +540,27 @@ __global__ __launch_bounds__(256) void mlp_layer_fwd_wide_f32(
   const float bias = jok ? B[j] : 0.f;
   const float* wrow = W + (long)j * in_d;
   const int in4 = in_d & ~3;
+  const int in8 = in_d & ~7;
   for (int rt = 0; rt < ROWS; rt += 16) {
+    // two independent accumulator chains double the MFMA/load ILP
+    // (the K chain is otherwise a serial dependent sequence)
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    // every lane of a 16-lane group loads the same 4-float W chunk
-    // (hardware broadcast) and selects its K element: 4x fewer load
-    // instructions than per-step scalar reads, fully pipelineable
-    for (int k0 = 0; k0 < in4; k0 += 4) {
-      float w0 = 0.f, w1 = 0.f, w2 = 0.f, w3 = 0.f;
+    f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < in8; k0 += 8) {
+      float bv0 = 0.f, bv1 = 0.f;
       if (jok) {
-        w0 = wrow[k0 + 0];
-        w1 = wrow[k0 + 1];
-        w2 = wrow[k0 + 2];
-        w3 = wrow[k0 + 3];
+        // every lane of a 16-lane group reads its K element of the
+        // broadcast W chunk (4x fewer load instructions than scalar)
+        bv0 = wrow[k0 + k];
+        bv1 = wrow[k0 + 4 + k];
       }
-      const float bv = k == 0 ? w0 : (k == 1 ? w1 : (k == 2 ? w2 : w3));
+      const float a0 = smem[(rt + i) * LDSW + k0 + k];
+      const float a1 = smem[(rt + i) * LDSW + k0 + 4 + k];
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, bv0, acc, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, bv1, acc2, 0, 0, 0);
+    }
+    for (int k0 = in8; k0 < in4; k0 += 4) {
+      float bv = jok ? wrow[k0 + k] : 0.f;
       const float a = smem[(rt + i) * LDSW + k0 + k];
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
     }
@@ -563,6 +570,7 @@ __global__ __launch_bounds__(256) void mlp_layer_fwd_wide_f32(
       float bv = (jok && kk < in_d) ? wrow[kk] : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
     }
+    acc[0] += acc2[0]; acc[1] += acc2[1]; acc[2] += acc2[2]; acc[3] += acc2[3];
     if (jok) {
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
@@ -603,14 +611,25 @@ __global__ __launch_bounds__(256) void mlp_dgrad_wide_f32(
   const int k = lane >> 4;
   const int j = jt + i;
   const bool jok = j < in_d;
+  const int od8 = out_d & ~7;
   for (int rt = 0; rt < ROWS; rt += 16) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    for (int k0 = 0; k0 < out_d; k0 += 4) {
+    f32x4 acc2 = {0.f, 0.f, 0.f, 0.f};
+    for (int k0 = 0; k0 < od8; k0 += 8) {
+      const float a0 = smem[(rt + i) * LDSW + k0 + k];
+      const float a1 = smem[(rt + i) * LDSW + k0 + 4 + k];
+      float bv0 = jok ? W[(long)(k0 + k) * in_d + j] : 0.f;
+      float bv1 = jok ? W[(long)(k0 + 4 + k) * in_d + j] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, bv0, acc, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, bv1, acc2, 0, 0, 0);
+    }
+    for (int k0 = od8; k0 < out_d; k0 += 4) {
       const int kk = k0 + k;
       float a = (kk < out_d) ? smem[(rt + i) * LDSW + kk] : 0.f;
       float bv = (jok && kk < out_d) ? W[(long)kk * in_d + j] : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
     }
+    acc[0] += acc2[0]; acc[1] += acc2[1]; acc[2] += acc2[2]; acc[3] += acc2[3];
     if (jok) {
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
