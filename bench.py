@@ -79,9 +79,11 @@ def main() -> None:
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
     use_gpu = torch.cuda.is_available()
-    device = f"cuda:{local_rank}" if use_gpu else "cpu"
+    # modulo lets multi-rank smoke tests share one GPU (gloo backend)
+    dev_idx = local_rank % max(1, torch.cuda.device_count()) if use_gpu else 0
+    device = f"cuda:{dev_idx}" if use_gpu else "cpu"
     if use_gpu:
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(dev_idx)
 
     from rl_replicas_amd.parallel import enable_data_parallel, init_from_env
 

@@ -11,7 +11,8 @@ bypass torch autograd entirely: per iteration
 with identical semantics to the eager path (same update order, same
 KL early stop, torch-exact min/clamp tie gradients — see
 loss_kernels.hip).  The DP gradient all-reduce hook runs between
-backward and step exactly as in the eager path.
+backward and step exactly as in the eager path, and the hipGraph
+variants keep it eager between captured pre/post graphs.
 
 Falls back silently (returns False from `supported`) for policies that
 are not MLP-backed Gaussian/Categorical or are not on GPU.
@@ -112,9 +113,10 @@ def _old_logp(policy, kind: str, obs: Tensor, actions: Tensor) -> Tensor:
 # 80-iteration update loops.  The Adam step counter lives on device, so
 # replays advance optimizer state correctly; inputs (obs / actions /
 # advantages / old_logp / returns) are copied into capture-stable
-# buffers each epoch.  Graphs are used only when (a) the optimizer is
-# the FusedAdam (device-side state) and (b) the run is single-process
-# (collectives are not captured); otherwise the eager fused path runs.
+# buffers each epoch.  Graphs require the FusedAdam optimizer
+# (device-side state).  Under data parallelism the iteration splits
+# into pre/post graphs around the eager gradient all-reduce —
+# collectives themselves are never captured.
 # ---------------------------------------------------------------------------
 
 

@@ -35,9 +35,6 @@ void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
                          const float* W, float* dx, float* ws, long ws_stride,
                          int batch, int out_d, int in_d, int act,
                          hipStream_t stream);
-__global__ void mlp_grad_reduce_stage_f32(const float* ws, float* out,
-                                          int n_blocks, int chunk, long grand);
-__global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a);
 void launch_gaussian_loss(const float* mean, const float* actions,
                           const float* old_logp, const float* adv,
                           const float* log_std, float* dmean, float* partials,

@@ -784,31 +784,7 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
 }
 
 // ---------------------------------------------------------------------------
-// two-stage deterministic partial reduction
-// stage 1 (layer-blind): out[c][e] = sum_{p in chunk c} ws[p][e]
-// stage 2: dw/db[e] = fixed-order sum over the <=REDUCE_CHUNKS chunk rows
-// ---------------------------------------------------------------------------
-__global__ void mlp_grad_reduce_stage_f32(const float* __restrict__ ws,
-                                          float* __restrict__ out, int n_blocks,
-                                          int chunk, long grand) {
-  const int c = blockIdx.y;
-  const int p0 = c * chunk;
-  const int p1 = min(p0 + chunk, n_blocks);
-  for (long e = blockIdx.x * blockDim.x + threadIdx.x; e < grand;
-       e += (long)gridDim.x * blockDim.x) {
-    float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-    int p = p0;
-    for (; p + 3 < p1; p += 4) {
-      s0 += ws[(p + 0) * grand + e];
-      s1 += ws[(p + 1) * grand + e];
-      s2 += ws[(p + 2) * grand + e];
-      s3 += ws[(p + 3) * grand + e];
-    }
-    for (; p < p1; ++p) s0 += ws[p * grand + e];
-    out[c * grand + e] = (s0 + s1) + (s2 + s3);
-  }
-}
-
+// deterministic split-K partial reduction
 // one-pass variant: 256 threads = 64 elements x 4 partial-quarters,
 // combined through LDS in fixed order — one launch even at 125 partials
 __global__ __launch_bounds__(256) void mlp_grad_reduce_onepass_f32(ReduceAllArgs a) {
@@ -842,24 +818,5 @@ __global__ __launch_bounds__(256) void mlp_grad_reduce_onepass_f32(ReduceAllArgs
       else a.db[l][idx - a.wsize[l]] = tot;
     }
     __syncthreads();
-  }
-}
-
-__global__ void mlp_grad_reduce_all_f32(ReduceAllArgs a) {
-  int grand = 0;
-  int base[MLP_MAX_LAYERS];
-  for (int l = 0; l < a.n_layers; ++l) {
-    base[l] = grand;
-    grand += a.total[l];
-  }
-  for (int g = blockIdx.x * blockDim.x + threadIdx.x; g < grand;
-       g += gridDim.x * blockDim.x) {
-    int l = 0;
-    while (l + 1 < a.n_layers && g >= base[l + 1]) ++l;
-    const int idx = g - base[l];
-    float s = 0.f;
-    for (int p = 0; p < a.n_blocks; ++p) s += a.ws[p * a.stride + g];
-    if (idx < a.wsize[l]) a.dw[l][idx] = s;
-    else a.db[l][idx - a.wsize[l]] = s;
   }
 }

@@ -64,7 +64,9 @@ def init_from_env(backend: Optional[str] = None, timeout_s: float = 600.0) -> in
     if world_size <= 1 and "MASTER_ADDR" not in os.environ:
         return 0
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("RL_REPLICAS_AMD_DIST_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo"
+        )
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if torch.cuda.is_available():
         torch.cuda.set_device(local_rank)
