@@ -69,7 +69,8 @@ def init_from_env(backend: Optional[str] = None, timeout_s: float = 600.0) -> in
         )
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if torch.cuda.is_available():
-        torch.cuda.set_device(local_rank)
+        # modulo lets multi-rank smoke tests share one GPU (gloo backend)
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
     from datetime import timedelta
 
     dist.init_process_group(backend=backend, timeout=timedelta(seconds=timeout_s))
