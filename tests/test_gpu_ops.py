@@ -318,3 +318,59 @@ class TestBF16Compute:
         model.learn(num_epochs=3, batch_size=500, output_dir=str(tmp_path))
         for p in pnet.parameters():
             assert torch.isfinite(p).all()
+
+
+class TestShapeStress:
+    """Odd layer counts/widths/batches through the fused fwd+bwd paths."""
+
+    @pytest.mark.parametrize(
+        "sizes",
+        [
+            [7, 24, 3],            # odd dims
+            [5, 64, 1],            # 2-layer value-like
+            [17, 48, 48, 48, 5],   # 4 layers
+            [9, 32, 32, 32, 32, 2],  # 5 layers (MLP_MAX_LAYERS)
+            [3, 8, 2],             # tiny
+        ],
+    )
+    @pytest.mark.parametrize("batch", [1, 17, 500])
+    def test_fwd_bwd_match_autograd(self, ext, sizes, batch):
+        from rl_replicas_amd.networks import MLP
+
+        torch.manual_seed(0)
+        mlp_f = MLP(sizes).to("cuda")
+        mlp_e = MLP(sizes).to("cuda")
+        mlp_e.load_state_dict(mlp_f.state_dict())
+        x = torch.randn(batch, sizes[0], device="cuda", requires_grad=True)
+        xe = x.detach().clone().requires_grad_(True)
+        out_f = mlp_f(x)
+        out_e = mlp_e.network(xe)
+        torch.testing.assert_close(out_f, out_e, rtol=2e-5, atol=2e-5)
+        g = torch.randn_like(out_e)
+        out_f.backward(g)
+        out_e.backward(g)
+        torch.testing.assert_close(x.grad, xe.grad, rtol=1e-3, atol=1e-4)
+        for p_f, p_e in zip(mlp_f.parameters(), mlp_e.parameters()):
+            torch.testing.assert_close(p_f.grad, p_e.grad, rtol=1e-3, atol=1e-4)
+
+    def test_single_layer_value_backward(self, ext):
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+        torch.manual_seed(0)
+        mlp = MLP([11, 1]).to("cuda")  # single Linear layer, identity head
+        weights, biases, acts = _extract_layers(mlp)
+        obs = torch.randn(300, 11, device="cuda")
+        ret = torch.randn(300, device="cuda")
+        outs = ext.mlp_forward(obs, list(weights), list(biases), acts, True)
+        grads = ext.value_mlp_backward(obs, list(weights), list(biases),
+                                       list(outs[1:]), outs[0], acts, ret)
+        v = mlp.network(obs).squeeze(-1).detach().requires_grad_(False)
+        w = weights[0].detach().clone().requires_grad_(True)
+        b = biases[0].detach().clone().requires_grad_(True)
+        v2 = (obs @ w.t() + b).squeeze(-1)
+        loss = torch.nn.functional.mse_loss(v2, ret)
+        loss.backward()
+        torch.testing.assert_close(grads[-1][0], loss.detach(), rtol=1e-4, atol=1e-6)
+        torch.testing.assert_close(grads[1], w.grad, rtol=1e-3, atol=1e-5)
+        torch.testing.assert_close(grads[2], b.grad, rtol=1e-3, atol=1e-5)
