@@ -124,7 +124,6 @@ def test_reference_checkpoint_loads():
 def test_rl_replicas_drop_in_alias():
     """Reference-style imports work through the rl_replicas shim."""
     from rl_replicas.algorithms import PPO as AliasPPO
-    from rl_replicas.policies import CategoricalPolicy as AliasCat
     from rl_replicas_amd.algorithms import PPO
 
     assert AliasPPO is PPO

@@ -19,12 +19,6 @@ def ext():
     return ops._load_extension()
 
 
-def eager_mlp(sizes, acts_mods):
-    from rl_replicas_amd.networks import MLP
-
-    return MLP(sizes, *acts_mods) if acts_mods else MLP(sizes)
-
-
 class TestFusedMLPForward:
     @pytest.mark.parametrize(
         "sizes,acts",

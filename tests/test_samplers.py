@@ -47,10 +47,6 @@ def manual_rollout(num_samples: int):
     )
 
 
-class _SeededRandomPolicy(RandomPolicy):
-    """RandomPolicy whose action space RNG is seeded like the oracle's."""
-
-
 def test_batch_sampler_matches_golden_replay():
     env = envs.make("CartPole-v1")
     env.action_space.seed(ENV_SEED)
