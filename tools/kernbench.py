@@ -37,7 +37,7 @@ def main():
     B = 4000
     results = {}
 
-    for rows in (32, 64):
+    for rows in (16, 32, 64):
         os.environ["RL_REPLICAS_AMD_MLP_ROWS"] = str(rows)
         mlp = MLP([17, 64, 32, 6]).to(dev)
         from rl_replicas_amd.ops.fused_mlp import _extract_layers
