@@ -109,7 +109,7 @@ void pick_tile(int batch, int max_width, int* rows, int* maxw) {
     const char* e = getenv("RL_REPLICAS_AMD_MLP_ROWS");
     return e ? atoi(e) : 0;
   }();
-  if (env_rows == 32 || env_rows == 64) *rows = env_rows;
+  if (env_rows == 16 || env_rows == 32 || env_rows == 64) *rows = env_rows;
   if (*maxw == 256) *rows = 32;  // only <32,256> is instantiated
 }
 
