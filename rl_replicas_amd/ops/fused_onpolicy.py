@@ -508,12 +508,23 @@ def value_update(algo, obs: Tensor, returns: Tensor, num_iters: int) -> float:
 
 
 def value_supported(algo, obs: Tensor) -> bool:
+    """Gate for the fused value loop: the MSE-seeded backward requires a
+    narrow net with a 1-output identity head (and an LDS-fitting weight
+    image); anything else falls back to the autograd path."""
     from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.ops.fused_mlp import ACT_IDENTITY
 
     vf = algo.value_function
-    return (
-        obs.is_cuda
-        and ops.hip_available()
-        and isinstance(vf.network, MLP)
-        and _extract_layers(vf.network) is not None
-    )
+    if not (obs.is_cuda and ops.hip_available() and isinstance(vf.network, MLP)):
+        return False
+    layout = _extract_layers(vf.network)
+    if layout is None:
+        return False
+    weights, biases, acts = layout
+    if weights[-1].shape[0] != 1 or acts[-1] != ACT_IDENTITY:
+        return False
+    max_width = max(obs.shape[-1], max(w.shape[0] for w in weights))
+    if max_width > 64:
+        return False
+    whole_w = sum(w.shape[0] * (w.shape[1] + 1) for w in weights)
+    return (3 * 32 * 68 + whole_w) * 4 <= 100 * 1024

@@ -129,3 +129,32 @@ def test_gpu_determinism_same_seed(tmp_path):
     p2 = run(tmp_path / "b")
     for a, b in zip(p1, p2):
         assert torch.equal(a, b)
+
+
+def test_ppo_with_wide_value_net_falls_back(tmp_path):
+    """A 256-wide value net doesn't satisfy the fused value-backward
+    constraints — PPO must fall back to the autograd value path, not
+    crash (regression test for the value_supported gate)."""
+    import torch.nn as nn
+
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import GaussianPolicy
+    from rl_replicas_amd.samplers import VectorSampler
+    from rl_replicas_amd.value_function import ValueFunction
+
+    torch.manual_seed(0)
+    venv = envs.VectorEnv("HalfCheetah-v4", num_envs=10)
+    pnet = MLP([17, 64, 32, 6]).to(DEVICE)
+    log_std = nn.Parameter(-0.5 * torch.ones(6, device=DEVICE))
+    policy = GaussianPolicy(
+        pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+    )
+    vnet = MLP([17, 256, 256, 1]).to(DEVICE)  # wide: fused value loop unsupported
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+    model = PPO(policy, vf, venv, VectorSampler(venv, seed=0),
+                num_policy_gradients=3, num_value_gradients=3)
+    model.learn(num_epochs=2, batch_size=300, output_dir=str(tmp_path))
+    for p in vnet.parameters():
+        assert torch.isfinite(p).all()
