@@ -2,6 +2,7 @@ from .spaces import Box, Discrete, Space
 from .core import Env, EnvSpec, make, register, registered_ids
 from .classic import CartPoleEnv, PendulumEnv
 from .synthetic import MUJOCO_SHAPES, SyntheticEnv
+from .subproc import SubprocVectorEnv
 from .vector import SerialVectorEnv, VectorEnv
 
 __all__ = [
@@ -19,4 +20,5 @@ __all__ = [
     "MUJOCO_SHAPES",
     "VectorEnv",
     "SerialVectorEnv",
+    "SubprocVectorEnv",
 ]
