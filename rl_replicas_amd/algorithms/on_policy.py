@@ -99,15 +99,40 @@ class OnPolicyAlgorithm(AlgorithmBase):
 
     # ------------------------------------------------------------------
     def _prepare_batch(self, experience: Experience) -> Dict[str, Tensor]:
-        """Device-resident flat batch with advantages and returns."""
+        """Device-resident flat batch with advantages and returns.
+
+        On GPU the fp32 payload (obs | actions | rewards | last_obs) is
+        packed into one cached pinned buffer and crosses in a single
+        H2D copy; offsets/dones are two tiny transfers."""
         device = self.device
         flat = experience.to_flat_batch()
-        obs = torch.as_tensor(flat["observations"], dtype=torch.float32, device=device)
-        actions = torch.as_tensor(
-            np.asarray(flat["actions"], dtype=np.float32), device=device
-        )
-        rewards = torch.as_tensor(flat["rewards"], dtype=torch.float32, device=device)
-        last_obs = torch.as_tensor(flat["last_observations"], dtype=torch.float32, device=device)
+        obs_np = flat["observations"]
+        act_np = np.asarray(flat["actions"], dtype=np.float32)
+        rew_np = flat["rewards"]
+        last_np = flat["last_observations"]
+        if device.type == "cuda":
+            sizes = [a.size for a in (obs_np, act_np, rew_np, last_np)]
+            total = int(sum(sizes))
+            pinned = getattr(self, "_h2d_pinned", None)
+            if pinned is None or pinned.numel() < total:
+                pinned = torch.empty(total, dtype=torch.float32, pin_memory=True)
+                self._h2d_pinned = pinned
+            host = pinned[:total].numpy()
+            o = 0
+            for a in (obs_np, act_np, rew_np, last_np):
+                host[o : o + a.size] = a.reshape(-1)
+                o += a.size
+            dev_flat = pinned[:total].to(device, non_blocking=True)
+            s0, s1, s2, s3 = np.cumsum(sizes)
+            obs = dev_flat[:s0].view(obs_np.shape)
+            actions = dev_flat[s0:s1].view(act_np.shape)
+            rewards = dev_flat[s1:s2]
+            last_obs = dev_flat[s2:s3].view(last_np.shape)
+        else:
+            obs = torch.as_tensor(obs_np, dtype=torch.float32, device=device)
+            actions = torch.as_tensor(act_np, device=device)
+            rewards = torch.as_tensor(rew_np, dtype=torch.float32, device=device)
+            last_obs = torch.as_tensor(last_np, dtype=torch.float32, device=device)
         offsets = torch.as_tensor(flat["episode_offsets"], device=device)
         dones = torch.as_tensor(flat["episode_dones"], device=device)
 

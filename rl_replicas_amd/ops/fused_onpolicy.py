@@ -96,6 +96,31 @@ def _backward_and_step(policy, mlp, obs, grad_out, hidden, final_out, weights,
     policy.optimizer.step()
 
 
+def _fast_diagnostics(policy, kind: str, obs: Tensor, actions: Tensor):
+    """policy/avarage_entropy + policy/log_prob_std without the torch
+    distribution machinery (reference logs these pre-update,
+    ppo.py:163-170).  For the state-independent-sigma Gaussian the mean
+    entropy is CLOSED FORM (sum(log_std) + D/2*(1+log 2pi)); logp comes
+    from the fused logp kernel."""
+    import math
+
+    ext = ops._load_extension()
+    logp = _old_logp(policy, kind, obs, actions)
+    if kind == "gaussian":
+        d = policy.log_std.numel()
+        entropy = float(policy.log_std.detach().sum()) + 0.5 * d * (1.0 + math.log(2 * math.pi))
+        return {
+            "policy/avarage_entropy": entropy,
+            "policy/log_prob_std": float(torch.std(logp)),
+        }
+    # categorical: entropy needs the full distribution; logits fwd is
+    # already done inside _old_logp's fused forward — recompute cheaply
+    with torch.no_grad():
+        dist = policy(obs)
+        ent = float(torch.mean(dist.entropy()))
+    return {"policy/avarage_entropy": ent, "policy/log_prob_std": float(torch.std(logp))}
+
+
 def _old_logp(policy, kind: str, obs: Tensor, actions: Tensor) -> Tensor:
     ext = ops._load_extension()
     mlp = _mlp_of(policy)
@@ -373,7 +398,7 @@ def ppo_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[s
         actions_k = actions.contiguous().view(-1)
     advantages = advantages.contiguous()
 
-    diagnostics = algo._policy_diagnostics(obs, actions)
+    diagnostics = _fast_diagnostics(policy, kind, obs, actions_k)
 
     with torch.no_grad():
         old_logp = _old_logp(algo.old_policy, kind, obs, actions_k)
@@ -449,7 +474,7 @@ def vpg_update(algo, obs: Tensor, actions: Tensor, advantages: Tensor) -> Dict[s
         actions_k = actions.contiguous().view(-1)
     advantages = advantages.contiguous()
 
-    diagnostics = algo._policy_diagnostics(obs, actions)
+    diagnostics = _fast_diagnostics(policy, kind, obs, actions_k)
 
     out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
     dummy = torch.empty(0, device=obs.device)
