@@ -17,7 +17,7 @@ from __future__ import annotations
 
 import copy
 import logging
-from typing import Dict, List
+from typing import Dict
 
 import numpy as np
 import torch
@@ -31,7 +31,7 @@ from rl_replicas_amd.experience import Experience
 from rl_replicas_amd.policies import Policy
 from rl_replicas_amd.replay_buffer import ReplayBuffer
 from rl_replicas_amd.samplers import Sampler
-from rl_replicas_amd.utils import add_noise_to_get_action, polyak_average
+from rl_replicas_amd.utils import add_noise_to_get_action
 
 logger = logging.getLogger(__name__)
 

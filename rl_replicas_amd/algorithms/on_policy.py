@@ -24,7 +24,7 @@ the CPU-oracle tests):
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 import torch

@@ -20,7 +20,7 @@ from __future__ import annotations
 
 import multiprocessing as mp
 from multiprocessing import shared_memory
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Optional
 
 import numpy as np
 

@@ -170,25 +170,37 @@ def gae_advantages_and_returns(
 
 
 def _gae_reference(rewards, values, last_values, episode_offsets, episode_dones, gamma, gae_lambda):
-    """Plain-PyTorch oracle (per-episode backward recurrences)."""
-    advantages = torch.empty_like(rewards)
-    returns = torch.empty_like(rewards)
+    """CPU oracle: per-episode backward recurrences in numpy (identical
+    math to the kernel; plain float arithmetic instead of per-element
+    torch dispatch keeps the CPU path usable at 4000-row batches)."""
+    import numpy as np
+
+    r = rewards.detach().cpu().numpy().astype(np.float32)
+    v = values.detach().cpu().numpy().astype(np.float32)
+    lv = last_values.detach().cpu().numpy().astype(np.float32)
     offs = episode_offsets.tolist()
-    n = len(offs) - 1
-    for e in range(n):
+    dones = [bool(d) for d in episode_dones]
+    adv = np.empty_like(r)
+    ret = np.empty_like(r)
+    gl = np.float32(gamma * gae_lambda)
+    g = np.float32(gamma)
+    for e in range(len(offs) - 1):
         lo, hi = offs[e], offs[e + 1]
-        last_v = last_values[e]
-        boot = torch.zeros((), dtype=rewards.dtype, device=rewards.device) if bool(episode_dones[e]) else last_v
-        ret_acc = boot
-        adv_acc = torch.zeros((), dtype=rewards.dtype, device=rewards.device)
+        last_v = lv[e]
+        ret_acc = np.float32(0.0) if dones[e] else last_v
+        adv_acc = np.float32(0.0)
         for t in range(hi - 1, lo - 1, -1):
-            ret_acc = rewards[t] + gamma * ret_acc
-            returns[t] = ret_acc
-            v_next = values[t + 1] if t + 1 < hi else last_v
-            delta = rewards[t] + gamma * v_next - values[t]
-            adv_acc = delta + gamma * gae_lambda * adv_acc
-            advantages[t] = adv_acc
-    return advantages, returns
+            ret_acc = r[t] + g * ret_acc
+            ret[t] = ret_acc
+            v_next = v[t + 1] if t + 1 < hi else last_v
+            delta = r[t] + g * v_next - v[t]
+            adv_acc = delta + gl * adv_acc
+            adv[t] = adv_acc
+    device = rewards.device
+    return (
+        torch.as_tensor(adv, dtype=rewards.dtype, device=device),
+        torch.as_tensor(ret, dtype=rewards.dtype, device=device),
+    )
 
 
 # ---------------------------------------------------------------------------

@@ -17,7 +17,7 @@ collected loss tensors once per epoch.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+
 
 import torch
 from torch import Tensor

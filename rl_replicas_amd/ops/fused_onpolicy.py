@@ -20,7 +20,7 @@ are not MLP-backed Gaussian/Categorical or are not on GPU.
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch

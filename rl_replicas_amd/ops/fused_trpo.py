@@ -31,7 +31,7 @@ import torch
 from torch import Tensor
 
 from rl_replicas_amd import ops
-from rl_replicas_amd.ops.fused_mlp import ACT_IDENTITY, ACT_RELU, ACT_TANH, _extract_layers
+from rl_replicas_amd.ops.fused_mlp import ACT_RELU, ACT_TANH, _extract_layers
 
 
 def _act_grad(code: int, y: Tensor) -> Optional[Tensor]:
