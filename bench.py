@@ -31,12 +31,12 @@ import torch
 import torch.nn as nn
 
 
-def build_model(device: str, num_envs: int, seed: int):
+def build_model(device: str, num_envs: int, seed: int, env_mode: str = "device"):
     from rl_replicas_amd import envs, ops
     from rl_replicas_amd.algorithms import PPO
     from rl_replicas_amd.networks import MLP
     from rl_replicas_amd.policies import GaussianPolicy
-    from rl_replicas_amd.samplers import VectorSampler
+    from rl_replicas_amd.samplers import DeviceSampler, VectorSampler
     from rl_replicas_amd.value_function import ValueFunction
 
     obs_dim, act_dim = envs.MUJOCO_SHAPES["HalfCheetah-v4"]
@@ -50,8 +50,14 @@ def build_model(device: str, num_envs: int, seed: int):
     vnet = MLP([obs_dim, 64, 32, 1]).to(device)
     vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
 
-    venv = envs.VectorEnv("HalfCheetah-v4", num_envs=num_envs)
-    sampler = VectorSampler(venv, seed=seed)
+    if env_mode == "device":
+        # GPU-resident envs: rollout (policy fwd + Philox sample + env
+        # dynamics) never leaves the device (envs/device.py)
+        venv = envs.DeviceVectorEnv("HalfCheetah-v4", num_envs=num_envs, device=device)
+        sampler = DeviceSampler(venv, seed=seed)
+    else:
+        venv = envs.VectorEnv("HalfCheetah-v4", num_envs=num_envs)
+        sampler = VectorSampler(venv, seed=seed)
     model = PPO(policy, vf, venv, sampler)  # gamma .99, lambda .97, eps .2, 80/80
     return model, sampler
 
@@ -64,6 +70,13 @@ def main() -> None:
     parser.add_argument("--batch-per-gpu", type=int, default=4000)
     parser.add_argument("--num-envs", type=int, default=200)
     parser.add_argument("--phase-timing", action="store_true", help="print sample/train ms split (rank 0, stderr)")
+    parser.add_argument(
+        "--env",
+        choices=["device", "cpu"],
+        default="device",
+        help="env residency: GPU-resident synthetic envs (default; same "
+        "dynamics, rollout stays on device) or numpy CPU VectorEnv",
+    )
     parser.add_argument(
         "--dtype",
         choices=["fp32", "bf16"],
@@ -93,7 +106,7 @@ def main() -> None:
     torch.manual_seed(1234 + rank)
     np.random.seed(1234 + rank)
 
-    model, sampler = build_model(device, args.num_envs, seed=1234 + rank)
+    model, sampler = build_model(device, args.num_envs, seed=1234 + rank, env_mode=args.env)
     if world > 1:
         enable_data_parallel(model)
 
@@ -169,6 +182,7 @@ def main() -> None:
                 "seq_len": 1000,
                 "parallelism": f"dp{world}",
                 "num_envs_per_gpu": args.num_envs,
+                "env_residency": args.env,
                 "policy_grads_per_epoch": 80,
                 "value_grads_per_epoch": 80,
                 "note": "step = one PPO epoch (sample batch + full update); vs_baseline is the reference's implied ~946 env-steps/s serial CPU throughput (BASELINE.md)",

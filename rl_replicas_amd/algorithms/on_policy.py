@@ -106,6 +106,13 @@ class OnPolicyAlgorithm(AlgorithmBase):
         H2D copy; offsets/dones are two tiny transfers."""
         device = self.device
         flat = experience.to_flat_batch()
+        if isinstance(flat["observations"], Tensor):
+            # device-resident rollout (DeviceSampler): nothing to upload
+            obs = flat["observations"].to(device)
+            actions = flat["actions"].to(device)
+            rewards = flat["rewards"].to(device)
+            last_obs = flat["last_observations"].to(device)
+            return self._finish_batch(flat, obs, actions, rewards, last_obs, device)
         obs_np = flat["observations"]
         act_np = np.asarray(flat["actions"], dtype=np.float32)
         rew_np = flat["rewards"]
@@ -133,6 +140,9 @@ class OnPolicyAlgorithm(AlgorithmBase):
             actions = torch.as_tensor(act_np, device=device)
             rewards = torch.as_tensor(rew_np, dtype=torch.float32, device=device)
             last_obs = torch.as_tensor(last_np, dtype=torch.float32, device=device)
+        return self._finish_batch(flat, obs, actions, rewards, last_obs, device)
+
+    def _finish_batch(self, flat, obs, actions, rewards, last_obs, device) -> Dict[str, Tensor]:
         offsets = torch.as_tensor(flat["episode_offsets"], device=device)
         dones = torch.as_tensor(flat["episode_dones"], device=device)
 

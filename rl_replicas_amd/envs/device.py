@@ -1,0 +1,109 @@
+"""GPU-resident vectorized synthetic environments.
+
+The CPU `VectorEnv` path steps numpy dynamics on the host, which makes
+every rollout step pay a D2H (actions) + H2D (observations) round trip
+plus host arithmetic.  For the synthetic MuJoCo-shaped benchmark envs
+(`envs/synthetic.py`) the dynamics are a few small matmuls — exactly
+the kind of work the GPU does for free next to the policy forward — so
+this module keeps the *entire* environment state in HBM (Isaac-Gym
+style): policy forward, Philox action sampling, env transition and
+reward all run on device and the epoch's rollout never touches the
+host.
+
+Semantics match `SyntheticEnv`/`VectorEnv` (same dynamics matrices —
+they are taken from the registered numpy env — same tanh/clip/reward
+formulas, same 1000-step horizon with autoreset); only the RNG streams
+differ (torch Philox here vs numpy PCG64 there), which is the same
+freedom the reference has across gymnasium versions.  Episodes advance
+in lockstep: every instance resets together, so truncation boundaries
+are host-known arithmetic and episode slicing costs nothing.
+
+No reference counterpart (the reference steps ONE gymnasium env on the
+CPU, batch_sampler.py:55-99); this is MI355X-first design, validated
+against the numpy env in tests/test_device_env.py.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor
+
+from .core import make
+from .synthetic import SyntheticEnv
+
+
+class DeviceVectorEnv:
+    """N synthetic env instances as device tensors, stepped in lockstep.
+
+    s' = tanh(s A + clip(a) B + noise*eps);  r = <s', w> - 0.1|a|^2
+    (identical to SyntheticEnv._step_b, synthetic.py:72-78).
+    """
+
+    def __init__(self, env: str, num_envs: int, device="cpu", **make_kwargs):
+        base = make(env, **make_kwargs)
+        if not isinstance(base, SyntheticEnv):
+            raise TypeError(
+                f"DeviceVectorEnv supports the synthetic benchmark envs (got {type(base).__name__})"
+            )
+        self.num_envs = int(num_envs)
+        self.device = torch.device(device)
+        self.observation_space = base.observation_space
+        self.action_space = base.action_space
+        self.spec = base.spec
+        self.noise = float(base.noise)
+        # identical dynamics to the numpy env (same seed-derived matrices)
+        self.A = torch.from_numpy(base.A).to(self.device)
+        self.B = torch.from_numpy(base.B).to(self.device)
+        self.w = torch.from_numpy(base.w).to(self.device)
+        self._gen = torch.Generator(device=self.device)
+        self._gen.manual_seed(torch.initial_seed() & 0x7FFFFFFFFFFFFFFF)
+        self.state: Optional[Tensor] = None
+        self._elapsed = 0  # lockstep: one counter for all instances
+
+    def seed(self, seed: Optional[int]) -> None:
+        if seed is not None:
+            self._gen.manual_seed(int(seed))
+            self.action_space.seed(seed + 1000)  # VectorEnv.seed contract
+
+    def _init_state(self) -> Tensor:
+        return 0.1 * torch.randn(
+            self.num_envs, self.A.shape[0], generator=self._gen, device=self.device
+        )
+
+    def reset(self, *, seed: Optional[int] = None) -> Tensor:
+        if seed is not None:
+            self.seed(seed)
+        self.state = self._init_state()
+        self._elapsed = 0
+        return self.state
+
+    def step(self, actions: Tensor) -> Tuple[Tensor, Tensor, bool, Tensor]:
+        """-> (obs [N,O], reward [N], truncated: host bool, final_obs [N,O]).
+
+        `truncated` is a scalar (lockstep horizon; synthetic envs never
+        terminate early).  On truncation all instances autoreset; `obs`
+        is post-reset, `final_obs` the true successor (GAE bootstrap).
+        """
+        a = actions.clamp(-1.0, 1.0)
+        eps = torch.randn(
+            self.num_envs, self.A.shape[0], generator=self._gen, device=self.device
+        )
+        state = torch.tanh(
+            torch.addmm(self.noise * eps, self.state, self.A).addmm_(a, self.B)
+        )
+        reward = state @ self.w - 0.1 * (a * a).sum(dim=1)
+        self._elapsed += 1
+        truncated = (
+            self.spec.max_episode_steps is not None
+            and self._elapsed >= self.spec.max_episode_steps
+        )
+        final_obs = state
+        if truncated:
+            state = self._init_state()
+            self._elapsed = 0
+        self.state = state
+        return state, reward, truncated, final_obs
+
+    def close(self) -> None:
+        pass

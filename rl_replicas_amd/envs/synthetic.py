@@ -16,6 +16,8 @@ exercises real, learnable structure.
 """
 from __future__ import annotations
 
+import zlib
+
 import numpy as np
 
 from .core import Env, EnvSpec, register
@@ -49,7 +51,9 @@ class SyntheticEnv(Env):
         self.action_space = Box(-1.0, 1.0, shape=(act_dim,), dtype=np.float32)
         self.spec = EnvSpec(id, max_episode_steps=max_episode_steps)
         self.noise = np.float32(noise)
-        rng = np.random.default_rng(abs(hash(id)) % (2**31))
+        # crc32, not hash(): str hash is per-process randomized in Python 3,
+        # which would give every rank/run different dynamics
+        rng = np.random.default_rng(zlib.crc32(id.encode()) % (2**31))
         A = rng.standard_normal((obs_dim, obs_dim))
         # scale A to spectral norm 0.95 for bounded dynamics
         s = np.linalg.svd(A, compute_uv=False)[0]

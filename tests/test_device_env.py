@@ -1,0 +1,170 @@
+"""DeviceVectorEnv / DeviceSampler: the GPU-resident rollout path.
+
+Runs on CPU (torch tensors on the cpu device) — the device path is
+backend-agnostic torch code; GPU coverage comes from test_gpu_train.py.
+"""
+import numpy as np
+import pytest
+import torch
+import torch.nn as nn
+
+from rl_replicas_amd import envs, ops
+from rl_replicas_amd.envs import DeviceVectorEnv
+from rl_replicas_amd.samplers import DeviceSampler
+
+
+def _policy(obs_dim=17, act_dim=6, seed=0):
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import GaussianPolicy
+
+    torch.manual_seed(seed)
+    net = MLP([obs_dim, 32, act_dim])
+    log_std = nn.Parameter(-0.5 * torch.ones(act_dim))
+    return GaussianPolicy(net, ops.make_adam(list(net.parameters()) + [log_std], lr=3e-4), log_std)
+
+
+class TestDeviceVectorEnv:
+    def test_dynamics_match_numpy_env_exactly(self):
+        """With noise=0 the torch env must reproduce the numpy env
+        (same seed-derived A/B/w, same tanh/clip/reward formulas)."""
+        n = 6
+        np_env = envs.VectorEnv("Walker2d-v4", num_envs=n, noise=0.0)
+        dev_env = DeviceVectorEnv("Walker2d-v4", num_envs=n, device="cpu", noise=0.0)
+        obs_np = np_env.reset(seed=0)
+        dev_env.reset(seed=0)
+        # force identical start states (the RNG streams legitimately differ)
+        dev_env.state = torch.from_numpy(obs_np.copy())
+        rng = np.random.default_rng(42)
+        for _ in range(10):
+            act = rng.uniform(-1.5, 1.5, size=(n, 6)).astype(np.float32)  # tests the clip
+            o_np, r_np, *_ = np_env.step(act.copy())
+            o_t, r_t, truncated, _ = dev_env.step(torch.from_numpy(act))
+            assert not truncated
+            np.testing.assert_allclose(o_t.numpy(), o_np, atol=1e-5)
+            np.testing.assert_allclose(r_t.numpy(), r_np, atol=1e-5)
+
+    def test_lockstep_truncation_and_autoreset(self):
+        env = DeviceVectorEnv("HalfCheetah-v4", num_envs=4, device="cpu", max_episode_steps=5)
+        env.reset(seed=1)
+        for t in range(5):
+            obs, reward, truncated, final_obs = env.step(torch.zeros(4, 6))
+            assert truncated == (t == 4)
+        # autoreset happened: live obs is a fresh init state, not final_obs
+        assert not torch.equal(obs, final_obs)
+        assert env._elapsed == 0
+        # next horizon runs again
+        for t in range(5):
+            _, _, truncated, _ = env.step(torch.zeros(4, 6))
+            assert truncated == (t == 4)
+
+    def test_seed_determinism(self):
+        outs = []
+        for _ in range(2):
+            env = DeviceVectorEnv("Hopper-v4", num_envs=3, device="cpu")
+            obs = env.reset(seed=9)
+            o, r, _, _ = env.step(torch.full((3, 3), 0.3))
+            outs.append((obs, o, r))
+        for a, b in zip(outs[0], outs[1]):
+            assert torch.equal(a, b)
+
+    def test_rejects_non_synthetic_envs(self):
+        with pytest.raises(TypeError):
+            DeviceVectorEnv("CartPole-v1", num_envs=2)
+
+
+class TestDeviceSampler:
+    def test_episode_structure_and_flat_cache(self):
+        env = DeviceVectorEnv("HalfCheetah-v4", num_envs=10, device="cpu", max_episode_steps=8)
+        sampler = DeviceSampler(env, seed=3, is_continuous=True)
+        policy = _policy()
+        exp = sampler.sample(200, policy)  # 20 steps/env: cuts at 8, 16, epoch end
+        assert sum(exp.episode_lengths) == 200
+        assert sorted(set(exp.episode_lengths)) == [4, 8]
+        # instance-major: every env contributes (8, True), (8, True), (4, False)
+        assert exp.episode_lengths[:3] == [8, 8, 4]
+        assert exp.episode_dones[:3] == [True, True, False]
+        assert len(exp.episode_returns) == 30
+
+        flat = exp.to_flat_batch()
+        assert isinstance(flat["observations"], torch.Tensor)
+        assert flat["observations"].shape == (200, 17)
+        assert flat["actions"].shape == (200, 6)
+        offs = flat["episode_offsets"]
+        assert offs.shape == (31,) and int(offs[-1]) == 200
+        # flat rewards sliced at the offsets reproduce the episode returns
+        rew = flat["rewards"]
+        for e in range(30):
+            seg = rew[int(offs[e]) : int(offs[e + 1])]
+            assert abs(float(seg.sum()) - exp.episode_returns[e]) < 1e-4
+
+    def test_boundary_parity_with_vector_sampler(self):
+        """Same horizon/epoch shape -> same episode length/done pattern as
+        the numpy VectorSampler (the semantics contract)."""
+        policy = _policy()
+        n, horizon = 5, 12
+        venv = envs.VectorEnv("HalfCheetah-v4", num_envs=n, max_episode_steps=horizon)
+        denv = DeviceVectorEnv("HalfCheetah-v4", num_envs=n, device="cpu", max_episode_steps=horizon)
+        from rl_replicas_amd.samplers import VectorSampler
+
+        vs = VectorSampler(venv, seed=0, is_continuous=True)
+        ds = DeviceSampler(denv, seed=0, is_continuous=True)
+        for _ in range(3):  # boundaries drift across epochs
+            e_v = vs.sample(100, policy)
+            e_d = ds.sample(100, policy)
+            assert e_v.episode_lengths == e_d.episode_lengths
+            assert e_v.episode_dones == e_d.episode_dones
+
+    def test_sampler_determinism(self):
+        def run():
+            env = DeviceVectorEnv("HalfCheetah-v4", num_envs=4, device="cpu")
+            sampler = DeviceSampler(env, seed=5)
+            exp = sampler.sample(40, _policy(seed=1))
+            return exp.to_flat_batch()["observations"].clone(), list(exp.episode_returns)
+
+        o1, r1 = run()
+        o2, r2 = run()
+        assert torch.equal(o1, o2)
+        assert r1 == r2
+
+    def test_prepare_batch_tensor_path_matches_numpy_path(self):
+        """_prepare_batch must produce identical advantages/returns from
+        the device flat cache and from an equivalent numpy cache."""
+        from rl_replicas_amd.algorithms import PPO
+        from rl_replicas_amd.experience import Experience
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.value_function import ValueFunction
+
+        env = DeviceVectorEnv("HalfCheetah-v4", num_envs=8, device="cpu", max_episode_steps=10)
+        sampler = DeviceSampler(env, seed=11)
+        policy = _policy(seed=2)
+        torch.manual_seed(3)
+        vnet = MLP([17, 32, 1])
+        vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+        model = PPO(policy, vf, env, sampler)
+
+        exp = sampler.sample(160, policy)
+        flat_t = exp.to_flat_batch()
+        batch_t = model._prepare_batch(exp)
+
+        exp_np = Experience()
+        exp_np.set_flat_cache({k: v.numpy() for k, v in flat_t.items()})
+        batch_np = model._prepare_batch(exp_np)
+
+        for key in ("advantages", "discounted_returns"):
+            torch.testing.assert_close(batch_t[key], batch_np[key], rtol=1e-5, atol=1e-5)
+
+    def test_ppo_end_to_end(self):
+        from rl_replicas_amd.algorithms import PPO
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.value_function import ValueFunction
+        import tempfile
+
+        env = DeviceVectorEnv("HalfCheetah-v4", num_envs=20, device="cpu", max_episode_steps=40)
+        sampler = DeviceSampler(env, seed=7, is_continuous=True)
+        policy = _policy(seed=4)
+        torch.manual_seed(5)
+        vnet = MLP([17, 64, 32, 1])
+        vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+        model = PPO(policy, vf, env, sampler)
+        model.learn(num_epochs=3, batch_size=400, output_dir=tempfile.mkdtemp())
+        assert model.current_total_steps == 1200

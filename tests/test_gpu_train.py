@@ -158,3 +158,36 @@ def test_ppo_with_wide_value_net_falls_back(tmp_path):
     model.learn(num_epochs=2, batch_size=300, output_dir=str(tmp_path))
     for p in vnet.parameters():
         assert torch.isfinite(p).all()
+
+
+def test_ppo_device_resident_rollout(tmp_path):
+    """GPU-resident env + DeviceSampler: the whole epoch (rollout and
+    update) runs on device; episodes/metrics stay consistent."""
+    import torch.nn as nn
+
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import GaussianPolicy
+    from rl_replicas_amd.samplers import DeviceSampler
+    from rl_replicas_amd.value_function import ValueFunction
+
+    torch.manual_seed(0)
+    denv = envs.DeviceVectorEnv(
+        "HalfCheetah-v4", num_envs=20, device=DEVICE, max_episode_steps=30
+    )
+    pnet = MLP([17, 64, 32, 6]).to(DEVICE)
+    log_std = nn.Parameter(-0.5 * torch.ones(6, device=DEVICE))
+    policy = GaussianPolicy(
+        pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+    )
+    vnet = MLP([17, 64, 32, 1]).to(DEVICE)
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+    sampler = DeviceSampler(denv, seed=3, is_continuous=True)
+    model = PPO(policy, vf, denv, sampler)
+    model.learn(num_epochs=3, batch_size=400, output_dir=str(tmp_path))
+    assert model.current_total_steps == 1200
+    flat = sampler.sample(400, policy).to_flat_batch()
+    assert flat["observations"].is_cuda and flat["rewards"].is_cuda
+    for p in list(pnet.parameters()) + list(vnet.parameters()):
+        assert torch.isfinite(p).all()
