@@ -60,11 +60,25 @@ class DeviceVectorEnv:
         self._gen.manual_seed(torch.initial_seed() & 0x7FFFFFFFFFFFFFFF)
         self.state: Optional[Tensor] = None
         self._elapsed = 0  # lockstep: one counter for all instances
+        # HIP fast path RNG: Philox keyed (seed ^ salt, offset).  The salt
+        # decorrelates the env stream from the policy's action-sampling
+        # stream, which is keyed on the same torch seed.
+        self._philox_seed = (torch.initial_seed() ^ 0x9E3779B97F4A7C15) & 0x7FFFFFFFFFFFFFFF
+        self._philox_offset = 0
 
     def seed(self, seed: Optional[int]) -> None:
         if seed is not None:
             self._gen.manual_seed(int(seed))
+            self._philox_seed = (int(seed) ^ 0x9E3779B97F4A7C15) & 0x7FFFFFFFFFFFFFFF
+            self._philox_offset = 0
             self.action_space.seed(seed + 1000)  # VectorEnv.seed contract
+
+    def _wants_hip(self) -> bool:
+        from rl_replicas_amd import ops
+
+        return self.device.type == "cuda" and ops.wants_hip(
+            self.state if self.state is not None else self.A
+        )
 
     def _init_state(self) -> Tensor:
         return 0.1 * torch.randn(
@@ -74,7 +88,17 @@ class DeviceVectorEnv:
     def reset(self, *, seed: Optional[int] = None) -> Tensor:
         if seed is not None:
             self.seed(seed)
-        self.state = self._init_state()
+        if self._wants_hip():
+            from rl_replicas_amd import ops
+
+            ext = ops._load_extension()
+            self.state = ext.synthetic_env_reset(
+                self.num_envs, self.A.shape[0], self.A,
+                self._philox_seed, self._philox_offset,
+            )
+            self._philox_offset += 1
+        else:
+            self.state = self._init_state()
         self._elapsed = 0
         return self.state
 
@@ -85,22 +109,35 @@ class DeviceVectorEnv:
         terminate early).  On truncation all instances autoreset; `obs`
         is post-reset, `final_obs` the true successor (GAE bootstrap).
         """
-        a = actions.clamp(-1.0, 1.0)
-        eps = torch.randn(
-            self.num_envs, self.A.shape[0], generator=self._gen, device=self.device
-        )
-        state = torch.tanh(
-            torch.addmm(self.noise * eps, self.state, self.A).addmm_(a, self.B)
-        )
-        reward = state @ self.w - 0.1 * (a * a).sum(dim=1)
         self._elapsed += 1
         truncated = (
             self.spec.max_episode_steps is not None
             and self._elapsed >= self.spec.max_episode_steps
         )
-        final_obs = state
+        if self._wants_hip():
+            from rl_replicas_amd import ops
+
+            ext = ops._load_extension()
+            # ONE kernel: clip + sA + aB + Philox noise + tanh + reward
+            # (+ the autoreset init states on lockstep horizon boundaries)
+            state, reward, final_obs = ext.synthetic_env_step(
+                self.state, actions.contiguous(), self.A, self.B, self.w,
+                self.noise, self._philox_seed, self._philox_offset, truncated,
+            )
+            self._philox_offset += 2
+        else:
+            a = actions.clamp(-1.0, 1.0)
+            eps = torch.randn(
+                self.num_envs, self.A.shape[0], generator=self._gen, device=self.device
+            )
+            state = torch.tanh(
+                torch.addmm(self.noise * eps, self.state, self.A).addmm_(a, self.B)
+            )
+            reward = state @ self.w - 0.1 * (a * a).sum(dim=1)
+            final_obs = state
+            if truncated:
+                state = self._init_state()
         if truncated:
-            state = self._init_state()
             self._elapsed = 0
         self.state = state
         return state, reward, truncated, final_obs

@@ -68,6 +68,15 @@ __global__ void gaussian_sample_kernel(const float* mean, const float* log_std,
 __global__ void categorical_sample_kernel(const float* logits, int64_t* out,
                                           int B, int N, uint64_t seed,
                                           uint64_t offset);
+__global__ void synthetic_env_step_kernel(const float* state, const float* actions,
+                                          const float* A, const float* Bm,
+                                          const float* w, float* s_out,
+                                          float* final_out, float* reward, int N,
+                                          int O, int Adim, float sigma,
+                                          uint64_t seed, uint64_t offset,
+                                          int do_reset);
+__global__ void synthetic_env_reset_kernel(float* s_out, int total,
+                                           uint64_t seed, uint64_t offset);
 __global__ void segmented_gae_kernel(const float* rewards, const float* values,
                                      const float* last_values, const int* offsets,
                                      const int* dones, float* advantages,
@@ -733,6 +742,45 @@ torch::Tensor categorical_sample(torch::Tensor logits, int64_t seed,
   return out;
 }
 
+std::vector<torch::Tensor> synthetic_env_step(torch::Tensor state,
+                                              torch::Tensor actions,
+                                              torch::Tensor A, torch::Tensor B,
+                                              torch::Tensor w, double sigma,
+                                              int64_t seed, int64_t offset,
+                                              bool do_reset) {
+  check_f32_gpu(state, "state");
+  check_f32_gpu(actions, "actions");
+  const int N = (int)state.size(0);
+  const int O = (int)state.size(1);
+  const int Adim = (int)actions.size(1);
+  TORCH_CHECK(O <= 64, "synthetic_env_step: obs_dim must be <= 64 (one wave per row)");
+  auto s_out = torch::empty_like(state);
+  auto final_out = do_reset ? torch::empty_like(state) : s_out;
+  auto reward = torch::empty({N}, state.options());
+  hipLaunchKernelGGL(synthetic_env_step_kernel, dim3(N), dim3(64), 0,
+                     current_stream(), state.data_ptr<float>(),
+                     actions.data_ptr<float>(), A.data_ptr<float>(),
+                     B.data_ptr<float>(), w.data_ptr<float>(),
+                     s_out.data_ptr<float>(), final_out.data_ptr<float>(),
+                     reward.data_ptr<float>(), N, O, Adim, (float)sigma,
+                     (uint64_t)seed, (uint64_t)offset, do_reset ? 1 : 0);
+  HIP_OK(hipGetLastError());
+  return {s_out, reward, final_out};
+}
+
+torch::Tensor synthetic_env_reset(int64_t num_envs, int64_t obs_dim,
+                                  torch::Tensor like, int64_t seed,
+                                  int64_t offset) {
+  auto s_out = torch::empty({num_envs, obs_dim}, like.options());
+  const int total = (int)(num_envs * obs_dim);
+  hipLaunchKernelGGL(synthetic_env_reset_kernel,
+                     dim3(std::min(256, (total + 255) / 256)), dim3(256), 0,
+                     current_stream(), s_out.data_ptr<float>(), total,
+                     (uint64_t)seed, (uint64_t)offset);
+  HIP_OK(hipGetLastError());
+  return s_out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlp_forward", &mlp_forward, "fused MLP forward (gfx950)",
         py::arg("x"), py::arg("weights"), py::arg("biases"), py::arg("acts"),
@@ -764,4 +812,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Philox Gaussian action sample (gfx950)");
   m.def("categorical_sample", &categorical_sample,
         "Philox categorical action sample (gfx950)");
+  m.def("synthetic_env_step", &synthetic_env_step,
+        "fused synthetic-env transition + reward (gfx950)");
+  m.def("synthetic_env_reset", &synthetic_env_reset,
+        "synthetic-env init states (gfx950)");
 }

@@ -24,6 +24,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "mlp_kernels.hip"),
         os.path.join(HIP_DIR, "loss_kernels.hip"),
         os.path.join(HIP_DIR, "sample_kernels.hip"),
+        os.path.join(HIP_DIR, "env_kernels.hip"),
         os.path.join(HIP_DIR, "rollout_kernels.hip"),
         os.path.join(HIP_DIR, "update_kernels.hip"),
     ],
