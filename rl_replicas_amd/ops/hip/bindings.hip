@@ -64,7 +64,8 @@ __global__ void value_mse_bwd_kernel(const float* v, const float* ret, float* dv
 __global__ void gaussian_sample_kernel(const float* mean, const float* log_std,
                                        float* out, int B, int D, uint64_t seed,
                                        uint64_t offset, float noise_scale,
-                                       float limit);
+                                       float limit,
+                                       const unsigned long long* offset_ptr);
 __global__ void categorical_sample_kernel(const float* logits, int64_t* out,
                                           int B, int N, uint64_t seed,
                                           uint64_t offset);
@@ -74,9 +75,13 @@ __global__ void synthetic_env_step_kernel(const float* state, const float* actio
                                           float* final_out, float* reward, int N,
                                           int O, int Adim, float sigma,
                                           uint64_t seed, uint64_t offset,
-                                          int do_reset);
+                                          int do_reset,
+                                          const unsigned long long* offset_ptr);
 __global__ void synthetic_env_reset_kernel(float* s_out, int total,
-                                           uint64_t seed, uint64_t offset);
+                                           uint64_t seed, uint64_t offset,
+                                           const unsigned long long* offset_ptr);
+__global__ void counter_add_kernel(unsigned long long* ctr,
+                                   unsigned long long delta);
 __global__ void segmented_gae_kernel(const float* rewards, const float* values,
                                      const float* last_values, const int* offsets,
                                      const int* dones, float* advantages,
@@ -709,20 +714,29 @@ void fused_polyak_(std::vector<torch::Tensor> srcs, std::vector<torch::Tensor> d
   }
 }
 
+static const unsigned long long* ctr_ptr(const c10::optional<torch::Tensor>& t) {
+  if (!t.has_value()) return nullptr;
+  TORCH_CHECK(t->scalar_type() == torch::kInt64 && t->is_cuda() && t->numel() == 1,
+              "offset_ctr must be a 1-element int64 CUDA tensor");
+  return reinterpret_cast<const unsigned long long*>(t->data_ptr<int64_t>());
+}
+
 torch::Tensor gaussian_sample(torch::Tensor mean, torch::Tensor log_std,
                               int64_t seed, int64_t offset, double noise_scale,
-                              double limit) {
+                              double limit,
+                              c10::optional<torch::Tensor> offset_ctr,
+                              c10::optional<torch::Tensor> out_opt) {
   check_f32_gpu(mean, "mean");
   const int B = (int)mean.size(0);
   const int D = (int)mean.size(1);
-  auto out = torch::empty_like(mean);
+  auto out = out_opt.has_value() ? *out_opt : torch::empty_like(mean);
   const int total = B * D;
   hipLaunchKernelGGL(gaussian_sample_kernel,
                      dim3(std::min(256, (total + 255) / 256)), dim3(256), 0,
                      current_stream(), mean.data_ptr<float>(),
                      log_std.data_ptr<float>(), out.data_ptr<float>(), B, D,
                      (uint64_t)seed, (uint64_t)offset, (float)noise_scale,
-                     (float)limit);
+                     (float)limit, ctr_ptr(offset_ctr));
   HIP_OK(hipGetLastError());
   return out;
 }
@@ -747,38 +761,58 @@ std::vector<torch::Tensor> synthetic_env_step(torch::Tensor state,
                                               torch::Tensor A, torch::Tensor B,
                                               torch::Tensor w, double sigma,
                                               int64_t seed, int64_t offset,
-                                              bool do_reset) {
+                                              bool do_reset,
+                                              c10::optional<torch::Tensor> offset_ctr,
+                                              c10::optional<torch::Tensor> s_out_opt,
+                                              c10::optional<torch::Tensor> final_out_opt,
+                                              c10::optional<torch::Tensor> reward_opt) {
   check_f32_gpu(state, "state");
   check_f32_gpu(actions, "actions");
   const int N = (int)state.size(0);
   const int O = (int)state.size(1);
   const int Adim = (int)actions.size(1);
   TORCH_CHECK(O <= 64, "synthetic_env_step: obs_dim must be <= 64 (one wave per row)");
-  auto s_out = torch::empty_like(state);
-  auto final_out = do_reset ? torch::empty_like(state) : s_out;
-  auto reward = torch::empty({N}, state.options());
+  auto s_out = s_out_opt.has_value() ? *s_out_opt : torch::empty_like(state);
+  auto final_out = final_out_opt.has_value() ? *final_out_opt
+                   : (do_reset ? torch::empty_like(state) : s_out);
+  auto reward = reward_opt.has_value() ? *reward_opt
+                                       : torch::empty({N}, state.options());
   hipLaunchKernelGGL(synthetic_env_step_kernel, dim3(N), dim3(64), 0,
                      current_stream(), state.data_ptr<float>(),
                      actions.data_ptr<float>(), A.data_ptr<float>(),
                      B.data_ptr<float>(), w.data_ptr<float>(),
                      s_out.data_ptr<float>(), final_out.data_ptr<float>(),
                      reward.data_ptr<float>(), N, O, Adim, (float)sigma,
-                     (uint64_t)seed, (uint64_t)offset, do_reset ? 1 : 0);
+                     (uint64_t)seed, (uint64_t)offset, do_reset ? 1 : 0,
+                     ctr_ptr(offset_ctr));
   HIP_OK(hipGetLastError());
   return {s_out, reward, final_out};
 }
 
 torch::Tensor synthetic_env_reset(int64_t num_envs, int64_t obs_dim,
                                   torch::Tensor like, int64_t seed,
-                                  int64_t offset) {
-  auto s_out = torch::empty({num_envs, obs_dim}, like.options());
+                                  int64_t offset,
+                                  c10::optional<torch::Tensor> offset_ctr,
+                                  c10::optional<torch::Tensor> out_opt) {
+  auto s_out = out_opt.has_value()
+                   ? *out_opt
+                   : torch::empty({num_envs, obs_dim}, like.options());
   const int total = (int)(num_envs * obs_dim);
   hipLaunchKernelGGL(synthetic_env_reset_kernel,
                      dim3(std::min(256, (total + 255) / 256)), dim3(256), 0,
                      current_stream(), s_out.data_ptr<float>(), total,
-                     (uint64_t)seed, (uint64_t)offset);
+                     (uint64_t)seed, (uint64_t)offset, ctr_ptr(offset_ctr));
   HIP_OK(hipGetLastError());
   return s_out;
+}
+
+void counter_add_(torch::Tensor ctr, int64_t delta) {
+  TORCH_CHECK(ctr.scalar_type() == torch::kInt64 && ctr.is_cuda() && ctr.numel() == 1,
+              "ctr must be a 1-element int64 CUDA tensor");
+  hipLaunchKernelGGL(counter_add_kernel, dim3(1), dim3(1), 0, current_stream(),
+                     reinterpret_cast<unsigned long long*>(ctr.data_ptr<int64_t>()),
+                     (unsigned long long)delta);
+  HIP_OK(hipGetLastError());
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -809,11 +843,23 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("categorical_kl", &categorical_kl, "approx KL for Categorical policy (gfx950)");
   m.def("value_mse_loss", &value_mse_loss, "fused value MSE fwd+bwd (gfx950)");
   m.def("gaussian_sample", &gaussian_sample,
-        "Philox Gaussian action sample (gfx950)");
+        "Philox Gaussian action sample (gfx950)", py::arg("mean"),
+        py::arg("log_std"), py::arg("seed"), py::arg("offset"),
+        py::arg("noise_scale"), py::arg("limit"),
+        py::arg("offset_ctr") = py::none(), py::arg("out") = py::none());
   m.def("categorical_sample", &categorical_sample,
         "Philox categorical action sample (gfx950)");
   m.def("synthetic_env_step", &synthetic_env_step,
-        "fused synthetic-env transition + reward (gfx950)");
+        "fused synthetic-env transition + reward (gfx950)", py::arg("state"),
+        py::arg("actions"), py::arg("A"), py::arg("B"), py::arg("w"),
+        py::arg("sigma"), py::arg("seed"), py::arg("offset"),
+        py::arg("do_reset"), py::arg("offset_ctr") = py::none(),
+        py::arg("s_out") = py::none(), py::arg("final_out") = py::none(),
+        py::arg("reward") = py::none());
   m.def("synthetic_env_reset", &synthetic_env_reset,
-        "synthetic-env init states (gfx950)");
+        "synthetic-env init states (gfx950)", py::arg("num_envs"),
+        py::arg("obs_dim"), py::arg("like"), py::arg("seed"), py::arg("offset"),
+        py::arg("offset_ctr") = py::none(), py::arg("out") = py::none());
+  m.def("counter_add_", &counter_add_,
+        "advance a device RNG counter (gfx950)");
 }

@@ -22,7 +22,9 @@ __global__ __launch_bounds__(64) void synthetic_env_step_kernel(
     const float* __restrict__ A, const float* __restrict__ Bm,
     const float* __restrict__ w, float* __restrict__ s_out,
     float* __restrict__ final_out, float* __restrict__ reward, int N, int O,
-    int Adim, float sigma, uint64_t seed, uint64_t offset, int do_reset) {
+    int Adim, float sigma, uint64_t seed, uint64_t offset, int do_reset,
+    const unsigned long long* __restrict__ offset_ptr) {
+  if (offset_ptr) offset += *offset_ptr;  // hipGraph-replay RNG advance
   const int row = blockIdx.x;
   if (row >= N) return;
   const float* s = state + (long)row * O;
@@ -65,7 +67,15 @@ __global__ __launch_bounds__(64) void synthetic_env_step_kernel(
 
 // fresh init states: s = 0.1 * eps  (SyntheticEnv._init_state)
 __global__ __launch_bounds__(256) void synthetic_env_reset_kernel(
-    float* __restrict__ s_out, int total, uint64_t seed, uint64_t offset) {
+    float* __restrict__ s_out, int total, uint64_t seed, uint64_t offset,
+    const unsigned long long* __restrict__ offset_ptr) {
+  if (offset_ptr) offset += *offset_ptr;
   for (int i = blockIdx.x * 256 + threadIdx.x; i < total; i += gridDim.x * 256)
     s_out[i] = 0.1f * philox_normal(seed, offset, (uint32_t)i);
+}
+
+// advance a device RNG counter (graph-replayable: one bump per epoch)
+__global__ void counter_add_kernel(unsigned long long* ctr,
+                                   unsigned long long delta) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *ctr += delta;
 }

@@ -15,7 +15,12 @@
 __global__ __launch_bounds__(256) void gaussian_sample_kernel(
     const float* __restrict__ mean, const float* __restrict__ log_std,
     float* __restrict__ out, int B, int D, uint64_t seed, uint64_t offset,
-    float noise_scale, float limit) {
+    float noise_scale, float limit,
+    const unsigned long long* __restrict__ offset_ptr) {
+  // offset_ptr: optional device counter added to `offset` -- lets a
+  // hipGraph replay draw fresh randomness (the by-value offset is
+  // frozen at capture; the counter advances via counter_add_kernel)
+  if (offset_ptr) offset += *offset_ptr;
   const int total = B * D;
   for (int i = blockIdx.x * 256 + threadIdx.x; i < total; i += gridDim.x * 256) {
     uint32_t r[4];

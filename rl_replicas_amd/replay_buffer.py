@@ -48,6 +48,16 @@ class ReplayBuffer:
 
     def add_experience(self, experience: Experience) -> None:
         """Append all transitions of `experience` to the ring."""
+        flat = getattr(experience, "_flat_cache", None)
+        if (
+            flat is not None
+            and isinstance(flat.get("observations"), torch.Tensor)
+            and "next_observations" in flat
+        ):
+            # device-resident rollout (DeviceSampler): tensor-to-tensor ring
+            # write, no host round trip
+            self._add_flat_tensors(flat)
+            return
         obs = np.asarray(np.stack(experience.flattened_observations), dtype=np.float32)
         if obs.ndim == 1:
             obs = obs[:, None]
@@ -78,6 +88,29 @@ class ReplayBuffer:
         first = min(n, self.buffer_size - pos)
         for key, arr in batch.items():
             t = torch.as_tensor(arr).to(dev, non_blocking=True)
+            self._storage[key][pos : pos + first] = t[:first]
+            if first < n:
+                self._storage[key][: n - first] = t[first:]
+        self._write = (pos + n) % self.buffer_size
+        self.current_size = min(self.current_size + n, self.buffer_size)
+
+    def _add_flat_tensors(self, flat: Dict[str, torch.Tensor]) -> None:
+        batch = {
+            "observations": flat["observations"],
+            "actions": flat["actions"],
+            "rewards": flat["rewards"],
+            "next_observations": flat["next_observations"],
+            "dones": flat["step_dones"],
+        }
+        if self._storage is None:
+            self._allocate(batch["observations"].shape[1], batch["actions"].shape[1:])
+        assert self._storage is not None
+        n = batch["rewards"].shape[0]
+        dev = self.device or torch.device("cpu")
+        pos = self._write
+        first = min(n, self.buffer_size - pos)
+        for key, t in batch.items():
+            t = t.to(dev, non_blocking=True)
             self._storage[key][pos : pos + first] = t[:first]
             if first < n:
                 self._storage[key][: n - first] = t[first:]

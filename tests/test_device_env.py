@@ -168,3 +168,66 @@ class TestDeviceSampler:
         model = PPO(policy, vf, env, sampler)
         model.learn(num_epochs=3, batch_size=400, output_dir=tempfile.mkdtemp())
         assert model.current_total_steps == 1200
+
+
+class TestDeviceSamplerOffPolicy:
+    def test_replay_buffer_tensor_path(self):
+        """DeviceSampler flat cache feeds the replay ring directly:
+        next_observations are true successors and dones mark cuts."""
+        from rl_replicas_amd.replay_buffer import ReplayBuffer
+
+        env = DeviceVectorEnv("Hopper-v4", num_envs=5, device="cpu", max_episode_steps=4)
+        sampler = DeviceSampler(env, seed=2, is_continuous=True)
+        policy = _policy(obs_dim=11, act_dim=3, seed=0)
+        buf = ReplayBuffer(1000)
+        exp = sampler.sample(30, policy)  # 6 steps/env: cut at t=3
+        flat = exp.to_flat_batch()
+        buf.add_experience(exp)
+        assert len(buf) == 30
+        obs = buf._storage["observations"][:30]
+        nxt = buf._storage["next_observations"][:30]
+        dones = buf._storage["dones"][:30]
+        steps = 6
+        for i in range(5):  # instance-major rows
+            for t in range(steps - 1):
+                r = i * steps + t
+                if t == 3:  # cut: done, successor is the pre-reset state
+                    assert dones[r] == 1.0
+                    assert not torch.equal(nxt[r], obs[r + 1])
+                else:
+                    assert dones[r] == 0.0
+                    torch.testing.assert_close(nxt[r], obs[r + 1])
+        # flat cache and ring agree
+        torch.testing.assert_close(obs, flat["observations"])
+
+    def test_ddpg_end_to_end_with_device_sampler(self, tmp_path):
+        import torch.nn as nn
+
+        from rl_replicas_amd.algorithms import DDPG
+        from rl_replicas_amd.evaluator import Evaluator
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import DeterministicPolicy, RandomPolicy
+        from rl_replicas_amd.q_function import QFunction
+        from rl_replicas_amd.replay_buffer import ReplayBuffer
+
+        torch.manual_seed(0)
+        env = DeviceVectorEnv("Hopper-v4", num_envs=10, device="cpu",
+                              max_episode_steps=50)
+        pnet = MLP([11, 32, 3], activation_function=nn.ReLU,
+                   output_activation_function=nn.Tanh)
+        policy = DeterministicPolicy(pnet, torch.optim.Adam(pnet.parameters(), lr=1e-3))
+        qnet = MLP([14, 32, 1], activation_function=nn.ReLU)
+        q = QFunction(qnet, torch.optim.Adam(qnet.parameters(), lr=1e-3))
+        model = DDPG(
+            policy, RandomPolicy(env.action_space), q, env,
+            DeviceSampler(env, seed=3, is_continuous=True),
+            ReplayBuffer(10000), Evaluator(seed=4),
+        )
+        model.learn(
+            num_epochs=8, batch_size=50, num_start_steps=100,
+            num_steps_before_update=100, num_train_steps=5,
+            num_evaluation_episodes=0, output_dir=str(tmp_path),
+        )
+        assert model.current_total_steps == 400
+        for p in pnet.parameters():
+            assert torch.isfinite(p).all()
