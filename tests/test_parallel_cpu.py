@@ -88,7 +88,47 @@ def _worker_ppo_sync(rank: int, world: int, tmpdir: str):
     dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("worker", [_worker_allreduce, _worker_ppo_sync])
+def _worker_ppo_device_sampler(rank: int, world: int, tmpdir: str):
+    """DP over the device-resident rollout path (the bench/SCALE shape:
+    DeviceVectorEnv + DeviceSampler + Gaussian policy + fused Adam)."""
+    dist = _init(rank, world, tmpdir)
+    import torch.nn as nn
+
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.parallel import enable_data_parallel
+    from rl_replicas_amd.policies import GaussianPolicy
+    from rl_replicas_amd.samplers import DeviceSampler
+    from rl_replicas_amd.utils import set_seed_for_rank
+    from rl_replicas_amd.value_function import ValueFunction
+
+    set_seed_for_rank(0, rank)
+    denv = envs.DeviceVectorEnv("HalfCheetah-v4", num_envs=10, device="cpu",
+                                max_episode_steps=40)
+    pnet = MLP([17, 32, 6])
+    log_std = nn.Parameter(-0.5 * torch.ones(6))
+    policy = GaussianPolicy(
+        pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+    )
+    vnet = MLP([17, 32, 1])
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+    model = PPO(policy, vf, denv, DeviceSampler(denv, seed=100 + rank))
+    enable_data_parallel(model)
+    model.learn(num_epochs=2, batch_size=200, output_dir=os.path.join(tmpdir, "out"))
+
+    import torch.distributed as tdist
+
+    for p in list(policy.parameters()) + list(vf.parameters()):
+        ref = p.detach().clone()
+        tdist.broadcast(ref, src=0)
+        assert torch.equal(ref, p.detach()), "rank divergence detected"
+    dist.destroy_process_group()
+
+
+@pytest.mark.parametrize(
+    "worker", [_worker_allreduce, _worker_ppo_sync, _worker_ppo_device_sampler]
+)
 def test_two_rank_gloo(worker, tmp_path):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     mp.spawn(worker, args=(2, str(tmp_path)), nprocs=2, join=True)
