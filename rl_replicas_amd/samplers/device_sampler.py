@@ -82,7 +82,13 @@ class DeviceSampler(Sampler):
 
         from rl_replicas_amd.ops import fused_rollout
 
-        if fused_rollout.supported(policy, env):
+        # when horizon % steps != 0 the truncation pattern cycles through
+        # up to horizon/gcd values — cap the graph cache and serve unseen
+        # patterns eagerly instead of capturing without bound
+        graph_ok = fused_rollout.supported(policy, env) and (
+            (steps, cuts, reset_epoch) in self._graphs or len(self._graphs) < 8
+        )
+        if graph_ok:
             if first and self.seed is not None:
                 env.seed(self.seed)
             obs_buf, act_buf, rew_buf, finals = self._sample_graphed(
