@@ -231,3 +231,38 @@ class TestDeviceSamplerOffPolicy:
         assert model.current_total_steps == 400
         for p in pnet.parameters():
             assert torch.isfinite(p).all()
+
+    def test_td3_end_to_end_with_device_sampler(self, tmp_path):
+        import torch.nn as nn
+
+        from rl_replicas_amd.algorithms import TD3
+        from rl_replicas_amd.evaluator import Evaluator
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import DeterministicPolicy, RandomPolicy
+        from rl_replicas_amd.q_function import QFunction
+        from rl_replicas_amd.replay_buffer import ReplayBuffer
+
+        torch.manual_seed(1)
+        env = DeviceVectorEnv("Hopper-v4", num_envs=10, device="cpu",
+                              max_episode_steps=50)
+        pnet = MLP([11, 32, 3], activation_function=nn.ReLU,
+                   output_activation_function=nn.Tanh)
+        policy = DeterministicPolicy(pnet, torch.optim.Adam(pnet.parameters(), lr=1e-3))
+
+        def q():
+            qnet = MLP([14, 32, 1], activation_function=nn.ReLU)
+            return QFunction(qnet, torch.optim.Adam(qnet.parameters(), lr=1e-3))
+
+        model = TD3(
+            policy, RandomPolicy(env.action_space), q(), q(), env,
+            DeviceSampler(env, seed=6, is_continuous=True),
+            ReplayBuffer(10000), Evaluator(seed=7),
+        )
+        model.learn(
+            num_epochs=8, batch_size=50, num_start_steps=100,
+            num_steps_before_update=100, num_train_steps=5,
+            num_evaluation_episodes=0, output_dir=str(tmp_path),
+        )
+        assert model.current_total_steps == 400
+        for p in pnet.parameters():
+            assert torch.isfinite(p).all()
