@@ -266,3 +266,48 @@ class TestDeviceSamplerOffPolicy:
         assert model.current_total_steps == 400
         for p in pnet.parameters():
             assert torch.isfinite(p).all()
+
+
+class TestDeviceSamplerEdgeCases:
+    @staticmethod
+    def _pol(o, a, seed=0):
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import GaussianPolicy
+
+        torch.manual_seed(seed)
+        net = MLP([o, 8, a])
+        ls = nn.Parameter(-0.5 * torch.ones(a))
+        return GaussianPolicy(
+            net, ops.make_adam(list(net.parameters()) + [ls], lr=1e-3), ls
+        )
+
+    def test_horizon_one_every_step_truncates(self):
+        env = DeviceVectorEnv("Swimmer-v4", num_envs=3, device="cpu", max_episode_steps=1)
+        s = DeviceSampler(env, seed=1, is_continuous=True)
+        e = s.sample(12, self._pol(8, 2))
+        assert e.episode_lengths == [1] * 12
+        assert all(e.episode_dones)
+        f = e.to_flat_batch()
+        assert f["observations"].shape == (12, 8)
+        assert f["last_observations"].shape == (12, 8)
+
+    def test_single_env_single_step_epochs(self):
+        env = DeviceVectorEnv("Swimmer-v4", num_envs=1, device="cpu", max_episode_steps=5)
+        s = DeviceSampler(env, seed=2, is_continuous=True)
+        p = self._pol(8, 2, seed=1)
+        for _ in range(7):  # crosses the horizon one step at a time
+            e = s.sample(1, p)
+            assert sum(e.episode_lengths) == 1
+
+    def test_multiple_cuts_per_epoch(self):
+        env = DeviceVectorEnv("Swimmer-v4", num_envs=2, device="cpu", max_episode_steps=3)
+        s = DeviceSampler(env, seed=3)
+        e = s.sample(14, self._pol(8, 2, seed=2))  # 7 steps: cuts at 2, 5
+        assert e.episode_lengths == [3, 3, 1] * 2
+        assert e.episode_dones == [True, True, False] * 2
+        assert e.to_flat_batch()["step_dones"].view(2, 7)[:, [2, 5]].all()
+
+    def test_non_divisible_batch_raises(self):
+        env = DeviceVectorEnv("Swimmer-v4", num_envs=2, device="cpu")
+        with pytest.raises(ValueError):
+            DeviceSampler(env).sample(13, self._pol(8, 2))
