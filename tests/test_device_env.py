@@ -311,3 +311,33 @@ class TestDeviceSamplerEdgeCases:
         env = DeviceVectorEnv("Swimmer-v4", num_envs=2, device="cpu")
         with pytest.raises(ValueError):
             DeviceSampler(env).sample(13, self._pol(8, 2))
+
+
+@pytest.mark.parametrize("algo", ["trpo", "vpg"])
+def test_trpo_vpg_on_device_sampler(algo, tmp_path):
+    """All three on-policy algorithms run over the device rollout path
+    (PPO is covered above; TRPO exercises the CG/old-policy machinery
+    against device flat batches)."""
+    from rl_replicas_amd.algorithms import TRPO, VPG
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.optimizers import ConjugateGradientOptimizer
+    from rl_replicas_amd.policies import GaussianPolicy
+    from rl_replicas_amd.value_function import ValueFunction
+
+    torch.manual_seed(0)
+    env = DeviceVectorEnv("Hopper-v4", num_envs=10, device="cpu", max_episode_steps=40)
+    pnet = MLP([11, 32, 3])
+    log_std = nn.Parameter(-0.5 * torch.ones(3))
+    params = list(pnet.parameters()) + [log_std]
+    if algo == "trpo":
+        policy = GaussianPolicy(pnet, ConjugateGradientOptimizer(params), log_std)
+    else:
+        policy = GaussianPolicy(pnet, ops.make_adam(params, lr=3e-4), log_std)
+    vnet = MLP([11, 32, 1])
+    vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
+    cls = TRPO if algo == "trpo" else VPG
+    model = cls(policy, vf, env, DeviceSampler(env, seed=3), num_value_gradients=5)
+    model.learn(num_epochs=2, batch_size=200, output_dir=str(tmp_path))
+    assert model.current_total_steps == 400
+    for p in params:
+        assert torch.isfinite(p).all()
