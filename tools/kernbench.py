@@ -89,6 +89,17 @@ def main():
     results["segmented_gae"] = time_fn(lambda: ext.segmented_gae(rw, vals, lv, offs, dn, 0.99, 0.97))
     results["normalize"] = time_fn(lambda: ext.normalize(rw))
 
+    # device-env transition (bench shape: 200 envs, HalfCheetah dims)
+    st = torch.randn(200, 17, device=dev)
+    at = torch.randn(200, 6, device=dev)
+    A = torch.randn(17, 17, device=dev)
+    Bm = torch.randn(6, 17, device=dev)
+    w = torch.randn(17, device=dev)
+    results["env_step"] = time_fn(
+        lambda: ext.synthetic_env_step(st, at, A, Bm, w, 0.05, 1, 2, False)
+    )
+    results["env_reset"] = time_fn(lambda: ext.synthetic_env_reset(200, 17, st, 1, 2))
+
     # adam on the policy params
     mlp = MLP([17, 64, 32, 6]).to(dev)
     params = [p.data for p in mlp.parameters()]
