@@ -19,7 +19,11 @@ class FusedAdam(torch.optim.Adam):
     """torch.optim.Adam with a fused multi-tensor HIP step."""
 
     @torch.no_grad()
-    def step(self, closure=None):
+    def step(self, closure=None, *, step_delta: float = 0.0, do_bump: bool = True):
+        """step_delta/do_bump: captured-loop mode — iteration i of a
+        hipGraph-captured loop passes step_delta=i, do_bump=False and the
+        loop bumps once by num_iters at the end (bump_steps); bitwise
+        identical to per-iteration bumps for integer-valued fp32 steps."""
         loss = None
         if closure is not None:
             with torch.enable_grad():
@@ -64,5 +68,22 @@ class FusedAdam(torch.optim.Adam):
                 float(beta2),
                 float(group["eps"]),
                 float(group.get("weight_decay", 0.0)),
+                float(step_delta),
+                bool(do_bump),
             )
         return loss
+
+    @torch.no_grad()
+    def bump_steps(self, amount: float) -> None:
+        """Advance every step counter by `amount` (one kernel)."""
+        from rl_replicas_amd import ops
+
+        ext = ops._load_extension()
+        steps = [
+            self.state[p]["step"]
+            for group in self.param_groups
+            for p in group["params"]
+            if p in self.state and "step" in self.state[p]
+        ]
+        if steps:
+            ext.adam_bump_(steps, float(amount))

@@ -350,12 +350,33 @@ class _GraphedValueLoop:
             self.pre = _CapturedLoop(iter_pre, state)
             self.post = _CapturedLoop(lambda: vf.optimizer.step(), state)
         else:
+            # whole-loop graph: per-iteration loss finalizes and Adam step
+            # bumps are DEFERRED — iteration i bakes step_delta=i into its
+            # Adam launch and writes its loss partials into row i; one
+            # batched finalize + one bump-by-num_iters close the loop.
+            # Bitwise-identical to the per-iteration form (integer fp32
+            # steps; same serial partial-sum order per row).
+            fb = int(ext.value_loss_partials_blocks(obs0.shape[0]))
+            self.partials = torch.zeros(num_iters, fb, device=obs0.device)
+
+            def iter_deferred(i: int):
+                out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
+                grads = ext.value_mlp_backward(
+                    self.obs, list(weights), list(biases), list(hidden), out,
+                    acts, self.returns, ops.compute_bf16(), self.partials[i],
+                )
+                n = len(weights)
+                for w, dw in zip(weights, grads[1 : 1 + n]):
+                    w.grad = dw
+                for b, db in zip(biases, grads[1 + n : 1 + 2 * n]):
+                    b.grad = db
+
             def body():
-                losses = []
-                for _ in range(num_iters):
-                    losses.append(iter_pre())
-                    vf.optimizer.step()
-                return losses
+                for i in range(num_iters):
+                    iter_deferred(i)
+                    vf.optimizer.step(step_delta=float(i), do_bump=False)
+                vf.optimizer.bump_steps(float(num_iters))
+                return ext.value_loss_finalize(self.partials, fb)
 
             self.loop = _CapturedLoop(body, state)
 
@@ -372,7 +393,7 @@ class _GraphedValueLoop:
                 self.post.replay()
             return float(torch.cat(losses).mean())
         losses = self.loop.replay()
-        return float(torch.cat(losses).mean())
+        return float(losses.mean())
 
 
 def _get_cached_graph(algo, attr: str, key, builder):

@@ -59,6 +59,8 @@ __global__ void categorical_logp_kernel(const float* logits, const float* action
 __global__ void categorical_kl_kernel(const float* logits, const float* actions,
                                       const float* old_logp, float* out, int B,
                                       int N);
+__global__ void value_loss_finalize_rows(const float* partials, float* scalars,
+                                         int rows, int fb, int row_stride);
 __global__ void value_mse_bwd_kernel(const float* v, const float* ret, float* dv,
                                      float* scalars, int B);
 __global__ void gaussian_sample_kernel(const float* mean, const float* log_std,
@@ -525,7 +527,8 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
                                               torch::Tensor final_out,
                                               std::vector<int64_t> acts,
                                               torch::Tensor returns,
-                                              int64_t compute_bf16) {
+                                              int64_t compute_bf16,
+                                              c10::optional<torch::Tensor> partials_out) {
   const int L = (int)weights.size();
   check_f32_gpu(x, "x");
   check_f32_gpu(returns, "returns");
@@ -553,7 +556,13 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
 
   const int fb = (batch + 31) / 32;
   torch::Tensor ws = torch::empty({(int64_t)fb, grand}, opts);
-  torch::Tensor loss_partials = torch::empty({fb}, opts);
+  // partials_out: caller-owned slice -> finalize is deferred (the captured
+  // value loop batches all iterations' finalizes into one kernel)
+  const bool deferred = partials_out.has_value();
+  if (deferred)
+    TORCH_CHECK(partials_out->numel() >= fb && partials_out->is_contiguous(),
+                "partials_out too small");
+  torch::Tensor loss_partials = deferred ? *partials_out : torch::empty({fb}, opts);
   torch::Tensor scalars = torch::empty({1}, opts);
   std::vector<torch::Tensor> dws(L), dbs(L);
   MLPBwdArgs ba{};
@@ -592,11 +601,13 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   hipLaunchKernelGGL(mlp_grad_reduce_onepass_f32, dim3(rb), dim3(256), 0,
                      stream, ra);
   HIP_OK(hipGetLastError());
-  // loss = sum of the per-block partials (fixed order)
-  hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(1), 0, stream,
-                     loss_partials.data_ptr<float>(), nullptr,
-                     scalars.data_ptr<float>(), fb, 0);
-  HIP_OK(hipGetLastError());
+  if (!deferred) {
+    // loss = sum of the per-block partials (fixed order)
+    hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(1), 0, stream,
+                       loss_partials.data_ptr<float>(), nullptr,
+                       scalars.data_ptr<float>(), fb, 0);
+    HIP_OK(hipGetLastError());
+  }
 
   std::vector<torch::Tensor> out;
   out.push_back(dx);
@@ -661,7 +672,8 @@ void fused_adam_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
                  std::vector<torch::Tensor> exp_avgs,
                  std::vector<torch::Tensor> exp_avg_sqs,
                  std::vector<torch::Tensor> steps, double lr, double beta1,
-                 double beta2, double eps, double weight_decay) {
+                 double beta2, double eps, double weight_decay,
+                 double step_delta, bool do_bump) {
   size_t n = params.size();
   for (size_t base = 0; base < n; base += MT_MAX_TENSORS) {
     AdamArgs a{};
@@ -680,16 +692,50 @@ void fused_adam_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
     a.beta2 = (float)beta2;
     a.eps = (float)eps;
     a.weight_decay = (float)weight_decay;
+    a.step_delta = (float)step_delta;
     int max_chunks = 1;
     for (int i = 0; i < a.n_tensors; ++i)
       max_chunks = std::max(max_chunks, (a.numel[i] + 8191) / 8192);
     hipLaunchKernelGGL(fused_adam_kernel, dim3(a.n_tensors, max_chunks),
                        dim3(256), 0, current_stream(), a);
     HIP_OK(hipGetLastError());
+    if (do_bump) {
+      a.step_delta = 1.f;
+      hipLaunchKernelGGL(adam_step_bump_kernel, dim3(1), dim3(MT_MAX_TENSORS), 0,
+                         current_stream(), a);
+      HIP_OK(hipGetLastError());
+    }
+  }
+}
+
+// bump-only entry (captured value loop: ONE bump of +num_iters per replay)
+void adam_bump_(std::vector<torch::Tensor> steps, double amount) {
+  size_t n = steps.size();
+  for (size_t base = 0; base < n; base += MT_MAX_TENSORS) {
+    AdamArgs a{};
+    a.n_tensors = (int)std::min((size_t)MT_MAX_TENSORS, n - base);
+    for (int i = 0; i < a.n_tensors; ++i)
+      a.step[i] = steps[base + i].data_ptr<float>();
+    a.step_delta = (float)amount;
     hipLaunchKernelGGL(adam_step_bump_kernel, dim3(1), dim3(MT_MAX_TENSORS), 0,
                        current_stream(), a);
     HIP_OK(hipGetLastError());
   }
+}
+
+// deferred value-loss finalize over all iterations at once
+torch::Tensor value_loss_finalize(torch::Tensor partials, int64_t fb) {
+  check_f32_gpu(partials, "partials");
+  const int rows = (int)partials.size(0);
+  const int stride = (int)partials.size(1);
+  TORCH_CHECK((int)fb <= stride, "fb exceeds partials row stride");
+  auto scalars = torch::empty({rows}, partials.options());
+  hipLaunchKernelGGL(value_loss_finalize_rows, dim3((rows + 255) / 256),
+                     dim3(256), 0, current_stream(),
+                     partials.data_ptr<float>(), scalars.data_ptr<float>(),
+                     rows, (int)fb, stride);
+  HIP_OK(hipGetLastError());
+  return scalars;
 }
 
 void fused_polyak_(std::vector<torch::Tensor> srcs, std::vector<torch::Tensor> dsts,
@@ -827,11 +873,22 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "value-net backward with fused MSE seed (gfx950)", py::arg("x"),
         py::arg("weights"), py::arg("biases"), py::arg("hidden"),
         py::arg("final_out"), py::arg("acts"), py::arg("returns"),
-        py::arg("compute_bf16") = 0);
+        py::arg("compute_bf16") = 0, py::arg("partials_out") = py::none());
+  m.def("value_loss_partials_blocks",
+        [](int64_t batch) { return (batch + 31) / 32; },
+        "partials row length used by value_mlp_backward");
+  m.def("value_loss_finalize", &value_loss_finalize,
+        "batched deferred value-loss finalize (gfx950)");
+  m.def("adam_bump_", &adam_bump_, "bump Adam step counters (gfx950)");
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");
   m.def("normalize", &normalize, "fused mean/std normalize (gfx950)");
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");
-  m.def("fused_adam_", &fused_adam_, "fused multi-tensor Adam (gfx950)");
+  m.def("fused_adam_", &fused_adam_, "fused multi-tensor Adam (gfx950)",
+        py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
+        py::arg("exp_avg_sqs"), py::arg("steps"), py::arg("lr"),
+        py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
+        py::arg("weight_decay"), py::arg("step_delta") = 0.0,
+        py::arg("do_bump") = true);
   m.def("fused_polyak_", &fused_polyak_, "fused multi-tensor Polyak (gfx950)");
   m.def("gaussian_policy_loss", &gaussian_policy_loss,
         "fused Gaussian VPG/PPO loss fwd+bwd (gfx950)");

@@ -100,6 +100,10 @@ struct AdamArgs {
   int numel[MT_MAX_TENSORS];
   int n_tensors;
   float lr, beta1, beta2, eps, weight_decay;
+  // added to step[t] when computing bias correction (captured value loop
+  // bakes per-iteration deltas so ONE bump per graph suffices); also the
+  // bump amount for adam_step_bump_kernel
+  float step_delta;
 };
 
 struct PolyakArgs {

@@ -156,6 +156,20 @@ __global__ void loss_partials_finalize(const float* __restrict__ partials,
   else scalars[0] = s;
 }
 
+// batched row finalize for the captured value loop: one thread per row,
+// SAME serial summation order as loss_partials_finalize -> a deferred
+// finalize is bitwise-identical to 80 per-iteration launches.
+__global__ void value_loss_finalize_rows(const float* __restrict__ partials,
+                                         float* __restrict__ scalars, int rows,
+                                         int fb, int row_stride) {
+  const int r = blockIdx.x * 256 + threadIdx.x;
+  if (r >= rows) return;
+  const float* row = partials + (long)r * row_stride;
+  float s = 0.f;
+  for (int p = 0; p < fb; ++p) s += row[p];
+  scalars[r] = s;
+}
+
 // logp only (old-policy snapshot at epoch start)
 __global__ __launch_bounds__(LOSS_THREADS) void gaussian_logp_kernel(
     const float* __restrict__ mean, const float* __restrict__ actions,

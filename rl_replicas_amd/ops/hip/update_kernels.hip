@@ -23,7 +23,7 @@ __global__ __launch_bounds__(256) void fused_adam_kernel(AdamArgs a) {
   const int c0 = blockIdx.y * MT_CHUNK;
   if (c0 >= n) return;
   const int c1 = min(c0 + MT_CHUNK, n);
-  const float s = a.step[t][0] + 1.f;
+  const float s = a.step[t][0] + 1.f + a.step_delta;
   const float bc1 = 1.f - __powf(a.beta1, s);
   const float bc2 = 1.f - __powf(a.beta2, s);
 
@@ -48,7 +48,7 @@ __global__ __launch_bounds__(256) void fused_adam_kernel(AdamArgs a) {
 // after fused_adam_kernel on the same stream)
 __global__ void adam_step_bump_kernel(AdamArgs a) {
   const int t = threadIdx.x;
-  if (t < a.n_tensors) a.step[t][0] += 1.f;
+  if (t < a.n_tensors) a.step[t][0] += a.step_delta;  // 1.0 for a plain step
 }
 
 __global__ __launch_bounds__(256) void fused_polyak_kernel(PolyakArgs a) {
