@@ -2,7 +2,7 @@ from .spaces import Box, Discrete, Space
 from .core import Env, EnvSpec, make, register, registered_ids
 from .classic import CartPoleEnv, PendulumEnv
 from .synthetic import MUJOCO_SHAPES, SyntheticEnv
-from .device import DeviceVectorEnv
+from .device import DevicePendulumEnv, DeviceVectorEnv
 from .subproc import SubprocVectorEnv
 from .vector import SerialVectorEnv, VectorEnv
 
@@ -23,4 +23,5 @@ __all__ = [
     "SerialVectorEnv",
     "SubprocVectorEnv",
     "DeviceVectorEnv",
+    "DevicePendulumEnv",
 ]

@@ -144,3 +144,78 @@ class DeviceVectorEnv:
 
     def close(self) -> None:
         pass
+
+
+class DevicePendulumEnv:
+    """Pendulum-v1 as a lockstep device vector env (pure torch ops — the
+    identical code runs on CPU and GPU; no HIP kernel needed at N x 2
+    state sizes).  Dynamics/reward formulas match `classic.PendulumEnv`
+    exactly (fp64 state, fp32 obs); only the init RNG stream differs
+    (torch generator vs numpy).  Pendulum never terminates early, so the
+    lockstep-truncation contract of `DeviceSampler` holds.
+    """
+
+    def __init__(self, num_envs: int, device="cpu"):
+        from .classic import PendulumEnv
+
+        base = PendulumEnv()
+        self.num_envs = int(num_envs)
+        self.device = torch.device(device)
+        self.observation_space = base.observation_space
+        self.action_space = base.action_space
+        self.spec = base.spec
+        self._gen = torch.Generator(device=self.device)
+        self._gen.manual_seed(torch.initial_seed() & 0x7FFFFFFFFFFFFFFF)
+        self.state: Optional[Tensor] = None  # [N, 2] float64: theta, theta_dot
+        self._elapsed = 0
+
+    def seed(self, seed: Optional[int]) -> None:
+        if seed is not None:
+            self._gen.manual_seed(int(seed))
+            self.action_space.seed(seed + 1000)
+
+    def _init_state(self) -> Tensor:
+        u = torch.rand(self.num_envs, 2, generator=self._gen, device=self.device,
+                       dtype=torch.float64)
+        import math
+
+        return (2.0 * u - 1.0) * torch.tensor([math.pi, 1.0], device=self.device,
+                                              dtype=torch.float64)
+
+    def _obs(self) -> Tensor:
+        th, thdot = self.state[:, 0], self.state[:, 1]
+        return torch.stack([torch.cos(th), torch.sin(th), thdot], dim=1).float()
+
+    def reset(self, *, seed: Optional[int] = None) -> Tensor:
+        if seed is not None:
+            self.seed(seed)
+        self.state = self._init_state()
+        self._elapsed = 0
+        return self._obs()
+
+    def step(self, actions: Tensor) -> Tuple[Tensor, Tensor, bool, Tensor]:
+        import math
+
+        th, thdot = self.state[:, 0], self.state[:, 1]
+        u = actions.to(torch.float64).reshape(self.num_envs, -1)[:, 0].clamp(-2.0, 2.0)
+        ang = torch.remainder(th + math.pi, 2 * math.pi) - math.pi
+        costs = ang * ang + 0.1 * thdot * thdot + 0.001 * u * u
+        newthdot = (thdot + (15.0 * torch.sin(th) + 3.0 * u) * 0.05).clamp(-8.0, 8.0)
+        newth = th + newthdot * 0.05
+        self.state = torch.stack([newth, newthdot], dim=1)
+        self._elapsed += 1
+        truncated = (
+            self.spec.max_episode_steps is not None
+            and self._elapsed >= self.spec.max_episode_steps
+        )
+        final_obs = self._obs()
+        obs = final_obs
+        if truncated:
+            self.state = self._init_state()
+            self._elapsed = 0
+            obs = self._obs()
+        reward = (-costs).float()
+        return obs, reward, truncated, final_obs
+
+    def close(self) -> None:
+        pass

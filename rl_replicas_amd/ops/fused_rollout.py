@@ -35,7 +35,10 @@ from rl_replicas_amd.ops.fused_mlp import _extract_layers
 
 
 def supported(policy, env) -> bool:
+    from rl_replicas_amd.envs.device import DeviceVectorEnv
+
     if not (env.device.type == "cuda" and ops.hip_available()
+            and isinstance(env, DeviceVectorEnv)  # synthetic-dynamics kernels
             and fop._graphs_common(None)
             and fop._policy_kind(policy) == "gaussian"):
         return False

@@ -67,8 +67,8 @@ class DeviceSampler(Sampler):
                 f"num_samples ({num_samples}) must be divisible by num_envs ({N})"
             )
         steps = num_samples // N
-        obs_dim = int(env.A.shape[0])
-        act_dim = int(env.B.shape[0])
+        obs_dim = int(env.observation_space.shape[0])
+        act_dim = int(env.action_space.shape[0])
 
         first = self._obs is None
         reset_epoch = first or not self.is_continuous

@@ -341,3 +341,68 @@ def test_trpo_vpg_on_device_sampler(algo, tmp_path):
     assert model.current_total_steps == 400
     for p in params:
         assert torch.isfinite(p).all()
+
+
+class TestDevicePendulumEnv:
+    def test_dynamics_match_numpy_pendulum_exactly(self):
+        from rl_replicas_amd.envs import DevicePendulumEnv
+
+        n = 5
+        np_env = envs.VectorEnv("Pendulum-v1", num_envs=n)
+        dev_env = DevicePendulumEnv(num_envs=n, device="cpu")
+        np_env.reset(seed=0)
+        dev_env.reset(seed=0)
+        # force identical internal state (RNG streams legitimately differ)
+        dev_env.state = torch.from_numpy(np_env.env.state.copy())
+        rng = np.random.default_rng(1)
+        for _ in range(10):
+            act = rng.uniform(-3, 3, size=(n, 1)).astype(np.float32)  # tests clip
+            o_np, r_np, *_ = np_env.step(act.copy())
+            o_t, r_t, truncated, _ = dev_env.step(torch.from_numpy(act))
+            assert not truncated
+            np.testing.assert_allclose(o_t.numpy(), o_np, atol=1e-6)
+            np.testing.assert_allclose(r_t.numpy(), r_np, atol=1e-5)
+
+    def test_device_sampler_and_ddpg_on_pendulum(self, tmp_path):
+        import torch.nn as nn
+
+        from rl_replicas_amd.algorithms import DDPG
+        from rl_replicas_amd.envs import DevicePendulumEnv
+        from rl_replicas_amd.evaluator import Evaluator
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import DeterministicPolicy, RandomPolicy
+        from rl_replicas_amd.q_function import QFunction
+        from rl_replicas_amd.replay_buffer import ReplayBuffer
+
+        torch.manual_seed(0)
+        env = DevicePendulumEnv(num_envs=10, device="cpu")
+        exp = DeviceSampler(env, seed=2, is_continuous=True).sample(50, self._pol())
+        assert exp.to_flat_batch()["observations"].shape == (50, 3)
+
+        pnet = MLP([3, 32, 1], activation_function=nn.ReLU,
+                   output_activation_function=nn.Tanh)
+        policy = DeterministicPolicy(pnet, torch.optim.Adam(pnet.parameters(), lr=1e-3))
+        qnet = MLP([4, 32, 1], activation_function=nn.ReLU)
+        q = QFunction(qnet, torch.optim.Adam(qnet.parameters(), lr=1e-3))
+        model = DDPG(
+            policy, RandomPolicy(env.action_space), q, env,
+            DeviceSampler(env, seed=3, is_continuous=True),
+            ReplayBuffer(10000), Evaluator(seed=4),
+        )
+        model.learn(
+            num_epochs=6, batch_size=50, num_start_steps=100,
+            num_steps_before_update=100, num_train_steps=5,
+            num_evaluation_episodes=2, evaluation_interval=100,
+            output_dir=str(tmp_path),
+        )
+        assert model.current_total_steps == 300
+
+    @staticmethod
+    def _pol():
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.policies import GaussianPolicy
+
+        torch.manual_seed(9)
+        net = MLP([3, 16, 1])
+        ls = nn.Parameter(-0.5 * torch.ones(1))
+        return GaussianPolicy(net, ops.make_adam(list(net.parameters()) + [ls], lr=1e-3), ls)
