@@ -38,7 +38,18 @@ def pick_device(device: Optional[str]) -> str:
     return "cuda" if torch.cuda.is_available() else "cpu"
 
 
-def make_sampler(env_id: str, seed: int, num_envs: int, is_continuous: bool = False):
+def make_sampler(env_id: str, seed: int, num_envs: int, is_continuous: bool = False,
+                 env_mode: str = "cpu", device: str = "cpu"):
+    if env_mode == "device":
+        # GPU-resident rollout (envs/device.py); synthetic MuJoCo shapes
+        # and Pendulum have lockstep device implementations
+        from rl_replicas_amd.samplers import DeviceSampler
+
+        if env_id == "Pendulum-v1":
+            denv = envs.DevicePendulumEnv(num_envs=num_envs, device=device)
+        else:
+            denv = envs.DeviceVectorEnv(env_id, num_envs=num_envs, device=device)
+        return denv, DeviceSampler(denv, seed=seed, is_continuous=is_continuous)
     if num_envs > 1:
         venv = envs.VectorEnv(env_id, num_envs=num_envs)
         return venv, VectorSampler(venv, seed=seed, is_continuous=is_continuous)
@@ -46,10 +57,11 @@ def make_sampler(env_id: str, seed: int, num_envs: int, is_continuous: bool = Fa
     return env, BatchSampler(env, seed=seed, is_continuous=is_continuous)
 
 
-def build_on_policy(env_id: str, seed: int, device: Optional[str], num_envs: int, optimizer: str):
+def build_on_policy(env_id: str, seed: int, device: Optional[str], num_envs: int,
+                    optimizer: str, env_mode: str = "cpu"):
     set_seed_for_libraries(seed)
     dev = pick_device(device)
-    env, sampler = make_sampler(env_id, seed, num_envs)
+    env, sampler = make_sampler(env_id, seed, num_envs, env_mode=env_mode, device=dev)
     obs_dim = env.observation_space.shape[0]
 
     if hasattr(env.action_space, "n"):
@@ -77,10 +89,12 @@ def build_on_policy(env_id: str, seed: int, device: Optional[str], num_envs: int
     return env, sampler, policy, value_function
 
 
-def build_off_policy(env_id: str, seed: int, device: Optional[str], num_envs: int, twin: bool):
+def build_off_policy(env_id: str, seed: int, device: Optional[str], num_envs: int,
+                     twin: bool, env_mode: str = "cpu"):
     set_seed_for_libraries(seed)
     dev = pick_device(device)
-    env, sampler = make_sampler(env_id, seed, num_envs, is_continuous=True)
+    env, sampler = make_sampler(env_id, seed, num_envs, is_continuous=True,
+                                env_mode=env_mode, device=dev)
     obs_dim = env.observation_space.shape[0]
     act_dim = env.action_space.shape[0]
 

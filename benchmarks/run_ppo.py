@@ -8,10 +8,10 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 from common import build_on_policy  # noqa: E402
 
 
-def run_ppo(env_id, seed, outdir, device=None, num_envs=20, num_epochs=None):
+def run_ppo(env_id, seed, outdir, device=None, num_envs=20, num_epochs=None, env_mode='cpu'):
     from rl_replicas_amd.algorithms import PPO
 
-    env, sampler, policy, value_function = build_on_policy(env_id, seed, device, num_envs, "adam")
+    env, sampler, policy, value_function = build_on_policy(env_id, seed, device, num_envs, "adam", env_mode=env_mode)
     model = PPO(policy, value_function, env, sampler)
     model.learn(num_epochs=num_epochs or 750, batch_size=4000, output_dir=outdir)
 
@@ -24,5 +24,6 @@ if __name__ == "__main__":
     p.add_argument("--device", default=None)
     p.add_argument("--num-envs", type=int, default=20)
     p.add_argument("--num-epochs", type=int, default=None)
+    p.add_argument("--env-mode", choices=["cpu", "device"], default="cpu")
     a = p.parse_args()
-    run_ppo(a.env, a.seed, a.outdir, a.device, a.num_envs, a.num_epochs)
+    run_ppo(a.env, a.seed, a.outdir, a.device, a.num_envs, a.num_epochs, a.env_mode)

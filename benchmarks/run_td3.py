@@ -8,11 +8,11 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 from common import build_off_policy  # noqa: E402
 
 
-def run_td3(env_id, seed, outdir, device=None, num_envs=1, num_epochs=None):
+def run_td3(env_id, seed, outdir, device=None, num_envs=1, num_epochs=None, env_mode='cpu'):
     from rl_replicas_amd.algorithms import TD3
 
     env, sampler, policy, exploration, qs, buffer, evaluator = build_off_policy(
-        env_id, seed, device, num_envs, twin=True
+        env_id, seed, device, num_envs, twin=True, env_mode=env_mode
     )
     model = TD3(policy, exploration, qs[0], qs[1], env, sampler, buffer, evaluator)
     model.learn(num_epochs=num_epochs or 20000, batch_size=50, output_dir=outdir)
@@ -26,5 +26,6 @@ if __name__ == "__main__":
     p.add_argument("--device", default=None)
     p.add_argument("--num-envs", type=int, default=1)
     p.add_argument("--num-epochs", type=int, default=None)
+    p.add_argument("--env-mode", choices=["cpu", "device"], default="cpu")
     a = p.parse_args()
-    run_td3(a.env, a.seed, a.outdir, a.device, a.num_envs, a.num_epochs)
+    run_td3(a.env, a.seed, a.outdir, a.device, a.num_envs, a.num_epochs, a.env_mode)
