@@ -133,3 +133,45 @@ class TestNoisedPolicy:
         for _ in range(20):
             a = noised.get_action_numpy(np.random.randn(3).astype(np.float32))
             assert np.all(np.abs(a) <= 2.0)
+
+
+class TestSegmentedGaeBruteForce:
+    """Third independent derivation: the textbook double-sum definitions
+    A_t = sum_l (gamma*lam)^l * delta_{t+l} and R_t = sum_k gamma^k r_{t+k}
+    + gamma^{L-t} * bootstrap, evaluated directly in O(L^2)."""
+
+    def test_matches_double_sum_definition(self):
+        rng = np.random.default_rng(0)
+        gamma, lam = 0.99, 0.97
+        lengths = [5, 11, 2]
+        dones = [False, True, False]
+        r_eps = [rng.standard_normal(L).astype(np.float32) for L in lengths]
+        v_eps = [rng.standard_normal(L).astype(np.float32) for L in lengths]
+        lv = rng.standard_normal(len(lengths)).astype(np.float32)
+
+        exp_adv, exp_ret = [], []
+        for r, v, last, d in zip(r_eps, v_eps, lv, dones):
+            L = len(r)
+            v_full = np.concatenate([v, [last]])  # deltas always bootstrap
+            deltas = r + gamma * v_full[1:] - v_full[:-1]
+            boot = 0.0 if d else float(last)
+            for t in range(L):
+                exp_adv.append(sum((gamma * lam) ** l * deltas[t + l] for l in range(L - t)))
+                exp_ret.append(
+                    sum(gamma ** k * r[t + k] for k in range(L - t))
+                    + gamma ** (L - t) * boot
+                )
+
+        offsets = np.zeros(len(lengths) + 1, dtype=np.int64)
+        np.cumsum(lengths, out=offsets[1:])
+        adv, ret = ops.gae_advantages_and_returns(
+            torch.as_tensor(np.concatenate(r_eps)),
+            torch.as_tensor(np.concatenate(v_eps)),
+            torch.as_tensor(lv),
+            torch.as_tensor(offsets),
+            torch.as_tensor(np.asarray(dones)),
+            gamma,
+            lam,
+        )
+        np.testing.assert_allclose(adv.numpy(), exp_adv, rtol=2e-5, atol=2e-5)
+        np.testing.assert_allclose(ret.numpy(), exp_ret, rtol=2e-5, atol=2e-5)
