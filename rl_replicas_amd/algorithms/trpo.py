@@ -97,6 +97,12 @@ class TRPO(OnPolicyAlgorithm):
                 self.policy, observations,
                 self.policy.optimizer.hvp_damping_coefficient,
             )
+        # under DP every CG iteration and line-search evaluation must see
+        # the GLOBAL batch (mean over ranks) or replicas diverge
+        reduce_hook = (
+            self._reduce_scalar_mean if getattr(self, "_dp_enabled", False) else None
+        )
         self.policy.optimizer.step(
-            compute_surrogate_loss, compute_kl_constraint, fisher_vector_product=fvp
+            compute_surrogate_loss, compute_kl_constraint,
+            fisher_vector_product=fvp, reduce_hook=reduce_hook,
         )

@@ -75,6 +75,7 @@ class ConjugateGradientOptimizer(Optimizer):
         loss_function: Callable,
         kl_divergence_function: Callable,
         fisher_vector_product: Optional[Callable] = None,
+        reduce_hook: Optional[Callable[[Tensor], Tensor]] = None,
     ) -> None:  # type: ignore[override]
         """One trust-region update.
 
@@ -86,6 +87,14 @@ class ConjugateGradientOptimizer(Optimizer):
         callable replacing the double-backward FVP — TRPO supplies the
         Gauss-Newton form, exact at its evaluation point where
         policy == old_policy (algorithms/trpo.py).
+
+        `reduce_hook` (extension, data parallelism): applied to every
+        FVP output and every line-search loss/KL evaluation so that ALL
+        ranks solve the same global-batch trust-region problem and take
+        bitwise-identical steps.  The caller pre-reduces `.grad`; mean of
+        per-rank Hv == global-batch Hv at equal shard sizes, so with this
+        hook the N-rank step equals the full-batch step
+        (tests/test_parallel_cpu.py::_worker_trpo_sync).
         """
         params: List[Tensor] = [
             p for group in self.param_groups for p in group["params"] if p.grad is not None
@@ -95,6 +104,11 @@ class ConjugateGradientOptimizer(Optimizer):
         hvp = fisher_vector_product or self._make_fisher_vector_product(
             kl_divergence_function, params
         )
+        if reduce_hook is not None:
+            local_hvp, local_loss, local_kl = hvp, loss_function, kl_divergence_function
+            hvp = lambda v: reduce_hook(local_hvp(v))  # noqa: E731
+            loss_function = lambda: reduce_hook(local_loss())  # noqa: E731
+            kl_divergence_function = lambda: reduce_hook(local_kl())  # noqa: E731
         direction = self._conjugate_gradient(hvp, loss_grad)
         # NaN guard on the direction (reference :83)
         direction = torch.nan_to_num(direction, nan=0.0)
