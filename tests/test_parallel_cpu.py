@@ -207,10 +207,61 @@ def _worker_ddpg_sync(rank: int, world: int, tmpdir: str):
     dist.destroy_process_group()
 
 
+def _worker_td3_sync(rank: int, world: int, tmpdir: str):
+    """TD3 under DP: twin critics + delayed policy/target updates happen
+    on the same schedule on every rank (identical step counts), so
+    replicas stay bitwise identical."""
+    dist = _init(rank, world, tmpdir)
+    import torch.nn as nn
+
+    from rl_replicas_amd import envs
+    from rl_replicas_amd.algorithms import TD3
+    from rl_replicas_amd.evaluator import Evaluator
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.parallel import enable_data_parallel
+    from rl_replicas_amd.policies import DeterministicPolicy, RandomPolicy
+    from rl_replicas_amd.q_function import QFunction
+    from rl_replicas_amd.replay_buffer import ReplayBuffer
+    from rl_replicas_amd.samplers import BatchSampler
+    from rl_replicas_amd.utils import set_seed_for_rank
+
+    set_seed_for_rank(0, rank)
+    env = envs.make("Pendulum-v1")
+    pnet = MLP([3, 16, 1], activation_function=nn.ReLU, output_activation_function=nn.Tanh)
+    policy = DeterministicPolicy(pnet, torch.optim.Adam(pnet.parameters(), lr=1e-3))
+
+    def q():
+        qnet = MLP([4, 16, 1], activation_function=nn.ReLU)
+        return QFunction(qnet, torch.optim.Adam(qnet.parameters(), lr=1e-3))
+
+    model = TD3(
+        policy, RandomPolicy(env.action_space), q(), q(), env,
+        BatchSampler(env, seed=100 + rank, is_continuous=True),
+        ReplayBuffer(5000), Evaluator(seed=rank),
+    )
+    enable_data_parallel(model)
+    model.learn(
+        num_epochs=4, batch_size=50, num_start_steps=50,
+        num_steps_before_update=50, num_train_steps=5,
+        num_evaluation_episodes=0, output_dir=os.path.join(tmpdir, "out"),
+    )
+
+    import torch.distributed as tdist
+
+    mods = [policy, model.q_function_1, model.q_function_2,
+            model.target_policy, model.target_q_function_1, model.target_q_function_2]
+    for m in mods:
+        for p_ in m.parameters():
+            ref = p_.detach().clone()
+            tdist.broadcast(ref, src=0)
+            assert torch.equal(ref, p_.detach()), "rank divergence detected"
+    dist.destroy_process_group()
+
+
 @pytest.mark.parametrize(
     "worker",
     [_worker_allreduce, _worker_ppo_sync, _worker_ppo_device_sampler,
-     _worker_trpo_sync, _worker_ddpg_sync],
+     _worker_trpo_sync, _worker_ddpg_sync, _worker_td3_sync],
 )
 def test_two_rank_gloo(worker, tmp_path):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
