@@ -142,9 +142,32 @@ class _NoisedPolicy(Policy):
         self.action_noise_scale = action_noise_scale
         self.action_limit = float(np.asarray(action_space.high).reshape(-1)[0])
         self.action_size = int(action_space.shape[0])
+        self._sample_offset = 0
+        self._dummy_log_std: Optional[Tensor] = None
 
     def get_action_tensor(self, observation: Tensor) -> Tensor:
+        from rl_replicas_amd import ops
+
         action = self.base_policy.get_action_tensor(observation)
+        if (
+            action.dim() == 2
+            and action.dtype == torch.float32
+            and ops.wants_hip(action)
+        ):
+            # ONE Philox kernel (noise_scale overrides sigma, limit clips)
+            # instead of the randn+mul+add+clip torch chain — the DDPG/TD3
+            # rollout hot path (gaussian_sample_kernel, sample_kernels.hip)
+            ext = ops._load_extension()
+            if self._dummy_log_std is None or self._dummy_log_std.device != action.device:
+                self._dummy_log_std = torch.zeros(action.shape[1], device=action.device)
+            self._sample_offset += 1
+            # salt decorrelates this stream from the base policies' streams
+            seed = (torch.initial_seed() ^ 0x517CC1B727220A95) & 0x7FFFFFFFFFFFFFFF
+            return ext.gaussian_sample(
+                action.contiguous(), self._dummy_log_std, seed,
+                self._sample_offset, float(self.action_noise_scale),
+                self.action_limit,
+            )
         noise_shape = action.shape if action.dim() > 1 else (self.action_size,)
         action = action + self.action_noise_scale * torch.randn(
             noise_shape, device=action.device, dtype=action.dtype
