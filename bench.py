@@ -24,11 +24,41 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import numpy as np
 import torch
 import torch.nn as nn
+
+
+def _relaunch_distributed(gpus: int) -> int:
+    """Self-exec under torch.distributed.run when --gpus N>1 is invoked
+    without a torchrun environment (WORLD_SIZE unset).
+
+    The driver may call `python bench.py --gpus 8` directly; without
+    this, that command would silently run ONE rank.  One rank per GPU
+    over RCCL; rendezvous on 127.0.0.1 (container hostnames may not
+    resolve)."""
+    import socket
+    import subprocess
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    cmd = [
+        sys.executable,
+        "-m",
+        "torch.distributed.run",
+        "--nnodes=1",
+        f"--nproc-per-node={gpus}",
+        "--master-addr=127.0.0.1",
+        f"--master-port={port}",
+        os.path.abspath(__file__),
+    ] + sys.argv[1:]
+    env = dict(os.environ)
+    env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    return subprocess.run(cmd, env=env).returncode
 
 
 def build_model(device: str, num_envs: int, seed: int, env_mode: str = "device"):
@@ -65,8 +95,8 @@ def build_model(device: str, num_envs: int, seed: int, env_mode: str = "device")
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
-    parser.add_argument("--steps", type=int, default=10)
-    parser.add_argument("--warmup", type=int, default=3)
+    parser.add_argument("--steps", type=int, default=None, help="timed PPO epochs (default: 300 on GPU so the timed region is seconds long, 3 on CPU)")
+    parser.add_argument("--warmup", type=int, default=None, help="untimed warmup epochs (default: 20 on GPU, 1 on CPU)")
     parser.add_argument("--batch-per-gpu", type=int, default=4000)
     parser.add_argument("--num-envs", type=int, default=200)
     parser.add_argument("--phase-timing", action="store_true", help="print sample/train ms split (rank 0, stderr)")
@@ -87,11 +117,24 @@ def main() -> None:
     args = parser.parse_args()
     os.environ["RL_REPLICAS_AMD_COMPUTE_DTYPE"] = args.dtype
 
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        sys.exit(_relaunch_distributed(args.gpus))
+
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if args.gpus > 1 and world != args.gpus:
+        raise SystemExit(
+            f"bench.py: --gpus {args.gpus} but WORLD_SIZE={world}; refusing to "
+            f"report n_gpus={args.gpus} from {world} rank(s). Launch with "
+            f"torchrun --nproc-per-node {args.gpus} or let bench.py self-exec."
+        )
 
     use_gpu = torch.cuda.is_available()
+    if args.steps is None:
+        args.steps = 300 if use_gpu else 3
+    if args.warmup is None:
+        args.warmup = 20 if use_gpu else 1
     # modulo lets multi-rank smoke tests share one GPU (gloo backend)
     dev_idx = local_rank % max(1, torch.cuda.device_count()) if use_gpu else 0
     device = f"cuda:{dev_idx}" if use_gpu else "cpu"
@@ -116,6 +159,7 @@ def main() -> None:
     model.metrics_manager.stdout = False  # keep the JSON line clean
 
     phase_ms = {"sample": 0.0, "train": 0.0}
+    epoch_returns = []  # per-epoch mean episode return (learning evidence)
 
     def one_epoch():
         t0 = time.perf_counter()
@@ -129,6 +173,8 @@ def main() -> None:
             torch.cuda.synchronize()
         phase_ms["sample"] += (t1 - t0) * 1000.0
         phase_ms["train"] += (time.perf_counter() - t1) * 1000.0
+        if experience.episode_returns:
+            epoch_returns.append(float(np.mean(experience.episode_returns)))
 
     def barrier_sync():
         if world > 1:
@@ -155,14 +201,13 @@ def main() -> None:
     total_env_steps = world * args.batch_per_gpu * args.steps
     value = total_env_steps / elapsed
     if rank == 0 and args.phase_timing:
-        import sys
-
         n = args.warmup + args.steps
         print(
             f"phase ms/epoch: sample={phase_ms['sample']/n:.2f} train={phase_ms['train']/n:.2f}",
             file=sys.stderr,
         )
     if rank == 0:
+        n_epochs = args.warmup + args.steps
         result = {
             "metric": "env_steps_per_sec",
             "value": value,
@@ -176,6 +221,24 @@ def main() -> None:
             "vs_baseline": value / 946.0,
             "dtype": args.dtype,
             "data": "synthetic",
+            "timed_region_s": elapsed,
+            "phase_ms_per_epoch": (
+                {
+                    "sample": phase_ms["sample"] / n_epochs,
+                    "train": phase_ms["train"] / n_epochs,
+                }
+                if args.phase_timing
+                else None  # exact split needs --phase-timing (adds syncs)
+            ),
+            "avg_return": {
+                "first_epoch": epoch_returns[0] if epoch_returns else None,
+                "last_epoch": epoch_returns[-1] if epoch_returns else None,
+                "note": "mean episode return on the synthetic HalfCheetah-shaped env "
+                "(rank 0), first vs last bench epoch; real training happens inside "
+                "the timed region so the return rises. MuJoCo-comparable returns "
+                "are impossible offline (BASELINE.md); algorithm math is instead "
+                "pinned to the reference by tests/test_reference_equivalence.py.",
+            },
             "config": {
                 "model": "PPO HalfCheetah-v4 (policy MLP [17,64,32,6] tanh, value [17,64,32,1])",
                 "global_batch": world * args.batch_per_gpu,

@@ -5,6 +5,7 @@ from rl_replicas_amd.utils import set_seed_for_libraries
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: requires an MI355X GPU (run with -m gpu)")
+    config.addinivalue_line("markers", "slow: takes tens of seconds on CPU")
 
 
 def pytest_collection_modifyitems(config, items):
