@@ -124,12 +124,18 @@ class OnPolicyAlgorithm(AlgorithmBase):
             if pinned is None or pinned.numel() < total:
                 pinned = torch.empty(total, dtype=torch.float32, pin_memory=True)
                 self._h2d_pinned = pinned
+                self._h2d_event = torch.cuda.Event()
+            # the previous epoch's non_blocking copy must have drained
+            # before the host repacks the same pinned pages (don't rely
+            # on implicit sync from later metric readbacks)
+            self._h2d_event.synchronize()
             host = pinned[:total].numpy()
             o = 0
             for a in (obs_np, act_np, rew_np, last_np):
                 host[o : o + a.size] = a.reshape(-1)
                 o += a.size
             dev_flat = pinned[:total].to(device, non_blocking=True)
+            self._h2d_event.record()
             s0, s1, s2, s3 = np.cumsum(sizes)
             obs = dev_flat[:s0].view(obs_np.shape)
             actions = dev_flat[s0:s1].view(act_np.shape)

@@ -18,7 +18,6 @@ class Evaluator:
     def __init__(self, seed: Optional[int] = None, vectorized: bool = True):
         self.seed = seed
         self.vectorized = vectorized
-        self._seeded = False
 
     def evaluate(self, policy: Policy, env, num_episodes: int) -> Tuple[List[float], List[int]]:
         if self.vectorized and hasattr(env, "_reset_b"):
@@ -57,8 +56,10 @@ class Evaluator:
         from rl_replicas_amd.envs.vector import VectorEnv
 
         venv = VectorEnv(env, num_episodes)
-        obs = venv.reset(seed=self.seed if not self._seeded else None)
-        self._seeded = True
+        # reseed on EVERY call, like the reference's reset(seed=self.seed)
+        # (evaluator.py:30) and our own serial path: periodic evaluations
+        # start from the same seeded initial states each time
+        obs = venv.reset(seed=self.seed)
         returns = np.zeros(num_episodes)
         lengths = np.zeros(num_episodes, dtype=np.int64)
         finished = np.zeros(num_episodes, dtype=bool)

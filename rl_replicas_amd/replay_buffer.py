@@ -82,6 +82,11 @@ class ReplayBuffer:
             "dones": dones,
         }
         n = len(rews)
+        if n > self.buffer_size:
+            # keep only the newest buffer_size transitions, like the
+            # reference's front-trim (replay_buffer.py:40-49)
+            batch = {k: arr[-self.buffer_size :] for k, arr in batch.items()}
+            n = self.buffer_size
         dev = self.device or torch.device("cpu")
         pos = self._write
         # ring write, possibly wrapping
@@ -106,6 +111,9 @@ class ReplayBuffer:
             self._allocate(batch["observations"].shape[1], batch["actions"].shape[1:])
         assert self._storage is not None
         n = batch["rewards"].shape[0]
+        if n > self.buffer_size:
+            batch = {k: t[-self.buffer_size :] for k, t in batch.items()}
+            n = self.buffer_size
         dev = self.device or torch.device("cpu")
         pos = self._write
         first = min(n, self.buffer_size - pos)
