@@ -64,6 +64,13 @@ class DDPG(OffPolicyAlgorithm):
         the fused kernel path (ops.fused_offpolicy)."""
         from rl_replicas_amd.ops import fused_offpolicy as fop
 
+        if fop.graph_supported(self, minibatch_size):
+            # single-process GPU: whole loop = one captured hipGraph
+            self._record_offpolicy_metrics(
+                fop.graphed_epoch(self, num_train_steps, minibatch_size)
+            )
+            return
+
         policy_losses: List[Tensor] = []
         q_losses: List[Tensor] = []
         all_q_values: List[Tensor] = []

@@ -68,6 +68,18 @@ class OffPolicyAlgorithm(AlgorithmBase):
             param.requires_grad = False
 
     # ------------------------------------------------------------------
+    def _record_offpolicy_metrics(self, metrics: Dict[str, float]) -> None:
+        """Record a graphed-epoch metric dict with the reference's
+        tensorboard flags (ddpg.py:234-253, td3.py:265-299)."""
+        m = self.metrics_manager
+        for tag, value in metrics.items():
+            to_tb = tag.endswith("average_loss") or tag.endswith("avarage_q-value")
+            if to_tb:
+                m.record_scalar(tag, value, self.current_total_steps, tensorboard=True)
+            else:
+                m.record_scalar(tag, value)
+
+    # ------------------------------------------------------------------
     def learn(
         self,
         num_epochs: int = 2000,

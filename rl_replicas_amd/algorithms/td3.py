@@ -63,12 +63,21 @@ class TD3(OffPolicyAlgorithm):
         for q in (self.target_q_function_1, self.target_q_function_2):
             for param in q.network.parameters():
                 param.requires_grad = False
+        self._smooth_offset = 0  # Philox offset for the GPU smoothing kernel
 
     # ------------------------------------------------------------------
     def train(self, replay_buffer: ReplayBuffer, num_train_steps: int, minibatch_size: int) -> None:
         """Twin-critic minibatch loop (reference td3.py:214-263); losses
         stay device-resident, fused kernel steps on GPU."""
         from rl_replicas_amd.ops import fused_offpolicy as fop
+
+        if fop.graph_supported(self, minibatch_size):
+            # single-process GPU: the whole 50-iteration loop is ONE
+            # captured hipGraph replay (fused_offpolicy._GraphedOffPolicy)
+            self._record_offpolicy_metrics(
+                fop.graphed_epoch(self, num_train_steps, minibatch_size)
+            )
+            return
 
         policy_losses: List[Tensor] = []
         q1_losses: List[Tensor] = []
@@ -183,17 +192,32 @@ class TD3(OffPolicyAlgorithm):
         return loss.detach()
 
     def compute_targets(self, next_observations: Tensor, rewards: Tensor, dones: Tensor) -> Tensor:
-        """Smoothed min-twin target (reference td3.py:325-341)."""
+        """Smoothed min-twin target (reference td3.py:325-341).
+
+        GPU: smoothing noise+clip+limit is ONE Philox kernel
+        (ops.td3_smooth) and the min-twin bootstrap is one fused kernel
+        (ops.q_target_min2); CPU keeps the reference's torch.randn RNG
+        stream (tests/test_reference_equivalence.py)."""
+        from rl_replicas_amd.ops import fused_offpolicy as fop
+
         with torch.no_grad():
             next_actions = self.target_policy(next_observations)
-            epsilon = self.target_noise_scale * torch.randn_like(next_actions)
-            epsilon = torch.clamp(epsilon, -self.target_noise_clip, self.target_noise_clip)
             action_limit = float(np.asarray(self.env.action_space.high).reshape(-1)[0])
-            next_actions = torch.clamp(next_actions + epsilon, -action_limit, action_limit)
+            if ops.wants_hip(next_actions):
+                self._smooth_offset += 1
+                next_actions = ops.td3_smooth(
+                    next_actions, self.target_noise_scale,
+                    self.target_noise_clip, action_limit,
+                    fop._stream_seed(), self._smooth_offset,
+                )
+            else:
+                next_actions = ops.td3_smooth(
+                    next_actions, self.target_noise_scale,
+                    self.target_noise_clip, action_limit, 0, 0,
+                )
             q1 = self.target_q_function_1(next_observations, next_actions)
             q2 = self.target_q_function_2(next_observations, next_actions)
-            target_q = torch.min(q1, q2)
-        return ops.q_target(rewards, dones.float(), target_q, self.gamma)
+        return ops.q_target_min2(rewards, dones.float(), q1, q2, self.gamma)
 
     # ------------------------------------------------------------------
     def _checkpoint_dict(self, epoch: int) -> Dict:

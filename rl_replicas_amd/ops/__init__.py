@@ -238,3 +238,32 @@ def q_target(rewards: Tensor, dones: Tensor, q_next: Tensor, gamma: float) -> Te
         ext = _load_extension()
         return ext.q_target(rewards.contiguous(), dones.contiguous(), q_next.contiguous(), float(gamma))
     return rewards + gamma * (1.0 - dones) * q_next
+
+
+def q_target_min2(rewards: Tensor, dones: Tensor, q1: Tensor, q2: Tensor, gamma: float) -> Tensor:
+    """TD3 min-twin bootstrap target r + gamma*(1-d)*min(q1,q2)
+    (reference td3.py:335-341) in one kernel."""
+    if wants_hip(rewards):
+        ext = _load_extension()
+        return ext.q_target_min2(
+            rewards.contiguous(), dones.contiguous(), q1.contiguous(),
+            q2.contiguous(), float(gamma)
+        )
+    return rewards + gamma * (1.0 - dones) * torch.min(q1, q2)
+
+
+def td3_smooth(actions: Tensor, scale: float, clip: float, limit: float,
+               seed: int, offset: int, offset_ctr: Optional[Tensor] = None) -> Tensor:
+    """Target-policy smoothing clamp(a + clamp(scale*z, +-clip), +-limit)
+    (reference td3.py:325-341).  GPU: ONE Philox kernel (graph-replay
+    safe via offset_ctr).  CPU: the reference's torch.randn chain, so the
+    CPU path keeps the reference's RNG stream semantics."""
+    if wants_hip(actions):
+        ext = _load_extension()
+        return ext.td3_smooth(
+            actions.contiguous(), int(seed), int(offset), float(scale),
+            float(clip), float(limit), offset_ctr
+        )
+    epsilon = scale * torch.randn_like(actions)
+    epsilon = torch.clamp(epsilon, -clip, clip)
+    return torch.clamp(actions + epsilon, -limit, limit)

@@ -5,8 +5,14 @@ runs 50 minibatch iterations per epoch, each a full torch-autograd
 Q-step + delayed actor step.  On GPU this path replaces autograd with
 direct kernel calls built from the already-tested primitives:
 
-  q_step:       Q fwd (fused MLP over cat[obs,act]) -> value-MSE loss
-                kernel -> fused MLP backward -> fused Adam.
+  gather:       ONE Philox kernel draws indices and gathers the
+                minibatch from the HBM ring, emitting the critic input
+                [obs|act] pre-concatenated (replay_gather,
+                offpolicy_kernels.hip)
+  targets:      target-policy fwd -> [TD3] Philox smoothing kernel ->
+                target-Q fwd(s) -> fused (min-twin) bootstrap kernel
+  q_step:       Q fwd (fused MLP) -> value-MSE loss kernel -> fused MLP
+                backward -> fused Adam.
   policy_step:  actor fwd -> Q fwd -> dQ/d(input) via the MLP dgrad
                 chain (Q's weight grads are simply not applied,
                 mirroring the reference's requires_grad freeze) ->
@@ -14,16 +20,34 @@ direct kernel calls built from the already-tested primitives:
 
 No scalar leaves the device inside the loop; the algorithms read the
 collected loss tensors once per epoch.
+
+Single-process, the WHOLE `num_train_steps` loop is captured into one
+hipGraph (`_GraphedOffPolicy`) and replayed per epoch: iteration i
+bakes Adam step_delta=i (one bump at the end, like the captured value
+loop in fused_onpolicy.py) and draws its randomness through a device
+counter, so replays stay correct as the optimizer state advances and
+the ring fills.  Under data parallelism the loop stays eager-fused
+(the per-iteration gradient all-reduce cannot live inside a capture).
 """
 from __future__ import annotations
 
-
+import logging
+from typing import Dict, List, Optional
 
 import torch
 from torch import Tensor
 
 from rl_replicas_amd import ops
 from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+logger = logging.getLogger(__name__)
+
+# decorrelates this module's Philox streams from the rollout/sampler ones
+_SEED_SALT = 0x2545F4914F6CDD1D
+
+
+def _stream_seed() -> int:
+    return (torch.initial_seed() ^ _SEED_SALT) & 0x7FFFFFFFFFFFFFFF
 
 
 def supported(module, obs: Tensor) -> bool:
@@ -41,7 +65,8 @@ def supported(module, obs: Tensor) -> bool:
 def _fwd_saved(mlp, x: Tensor):
     ext = ops._load_extension()
     weights, biases, acts = _extract_layers(mlp)
-    outs = ext.mlp_forward(x, list(weights), list(biases), acts, True)
+    outs = ext.mlp_forward(x, list(weights), list(biases), acts, True,
+                           ops.compute_bf16())
     return outs[0], outs[1:], weights, biases, acts
 
 
@@ -49,7 +74,7 @@ def _backward(mlp, x, grad_out, hidden, final_out, weights, biases, acts):
     ext = ops._load_extension()
     return ext.mlp_backward(
         grad_out.contiguous(), x, list(weights), list(biases), list(hidden),
-        final_out, acts,
+        final_out, acts, ops.compute_bf16(),
     )
 
 
@@ -64,21 +89,31 @@ def _apply_grads(module, weights, biases, grads, extra=()):
 
 
 def q_step(q_function, observations: Tensor, actions: Tensor, targets: Tensor,
-           all_reduce_hook) -> Tensor:
-    """One critic MSE step; returns the loss as a device scalar."""
+           all_reduce_hook, qin: Optional[Tensor] = None,
+           step_delta: Optional[float] = None) -> Tensor:
+    """One critic MSE step; returns the loss as a device scalar.
+
+    `qin`: pre-concatenated [obs|act] (the fused gather emits it — skips
+    the torch.cat).  `step_delta`: captured-loop mode (deferred Adam
+    step-counter bump)."""
     ext = ops._load_extension()
-    qin = torch.cat([observations, actions], dim=-1).contiguous()
+    if qin is None:
+        qin = torch.cat([observations, actions], dim=-1).contiguous()
     mlp = q_function.network
     out, hidden, weights, biases, acts = _fwd_saved(mlp, qin)
     dv, scalars = ext.value_mse_loss(out.view(-1), targets.contiguous())
     grads = _backward(mlp, qin, dv.view(out.shape), hidden, out, weights, biases, acts)
     _apply_grads(q_function, weights, biases, grads)
     all_reduce_hook(q_function)
-    q_function.optimizer.step()
+    if step_delta is None:
+        q_function.optimizer.step()
+    else:
+        q_function.optimizer.step(step_delta=step_delta, do_bump=False)
     return scalars[0]
 
 
-def policy_step(policy, q_function, observations: Tensor, all_reduce_hook) -> Tensor:
+def policy_step(policy, q_function, observations: Tensor, all_reduce_hook,
+                step_delta: Optional[float] = None) -> Tensor:
     """One deterministic-actor step through a (frozen) critic:
     loss = -mean(Q(s, mu(s)))  (reference ddpg.py:255-273).
     Returns the loss as a device scalar."""
@@ -102,11 +137,210 @@ def policy_step(policy, q_function, observations: Tensor, all_reduce_hook) -> Te
     a_grads = _backward(pm, observations, d_act, a_hidden, a_out, a_w, a_b, a_acts)
     _apply_grads(policy, a_w, a_b, a_grads)
     all_reduce_hook(policy)
-    policy.optimizer.step()
+    if step_delta is None:
+        policy.optimizer.step()
+    else:
+        policy.optimizer.step(step_delta=step_delta, do_bump=False)
     return -q_out.mean()
 
 
 def forward_only(module, x: Tensor) -> Tensor:
     ext = ops._load_extension()
     weights, biases, acts = _extract_layers(module.network)
-    return ext.mlp_forward(x.contiguous(), list(weights), list(biases), acts, False)[0]
+    return ext.mlp_forward(x.contiguous(), list(weights), list(biases), acts,
+                           False, ops.compute_bf16())[0]
+
+
+# ---------------------------------------------------------------------------
+# hipGraph-captured epoch loop
+# ---------------------------------------------------------------------------
+
+
+def _noop_hook(_module) -> None:
+    pass
+
+
+def graph_supported(algo, minibatch_size: int) -> bool:
+    """Whole-loop capture needs: GPU ring buffer with storage allocated,
+    FusedAdam everywhere, fused-MLP-compatible nets, graphs enabled, and
+    no data parallelism (the per-iteration all-reduce stays eager)."""
+    import os
+
+    from rl_replicas_amd.ops.fused_adam import FusedAdam
+    from rl_replicas_amd.ops.fused_onpolicy import _dp_active
+
+    if os.environ.get("RL_REPLICAS_AMD_DISABLE_GRAPHS", "0") == "1":
+        return False
+    buf = algo.replay_buffer
+    if buf._storage is None or buf.device is None or buf.device.type != "cuda":
+        return False
+    if buf.current_size <= 0 or not ops.hip_available():
+        return False
+    if _dp_active(algo):
+        return False
+    obs0 = buf._storage["observations"][:1]
+    modules = [algo.policy, algo.target_policy]
+    if hasattr(algo, "q_function"):
+        modules += [algo.q_function, algo.target_q_function]
+    else:
+        modules += [
+            algo.q_function_1, algo.q_function_2,
+            algo.target_q_function_1, algo.target_q_function_2,
+        ]
+    for m in modules:
+        if not supported(m, obs0):
+            return False
+    for m in (algo.policy,) + tuple(
+        m for m in modules if hasattr(m, "optimizer") and m.optimizer is not None
+    ):
+        opt = getattr(m, "optimizer", None)
+        if opt is not None and not isinstance(opt, FusedAdam):
+            return False
+    return True
+
+
+class _GraphedOffPolicy:
+    """The whole num_train_steps DDPG/TD3 loop as ONE hipGraph.
+
+    Per iteration (reference td3.py:221-263 order):
+      replay_gather -> Q logging fwds -> target chain -> q_step(s)
+      [every policy_delay] policy_step + fused Polyak
+    RNG: iteration i uses Philox offsets (3i, 3i+1) plus a device
+    counter bumped by 3*num_iters inside the graph, so each replay
+    draws fresh minibatches/noise.  Adam bumps are deferred (step_delta
+    baked per iteration; one bump kernel per optimizer at the end) —
+    bitwise-identical to per-iteration stepping.
+    """
+
+    def __init__(self, algo, num_iters: int, minibatch_size: int):
+        from rl_replicas_amd.ops.fused_onpolicy import (
+            _CapturedLoop,
+            _ensure_adam_state,
+        )
+        from rl_replicas_amd.ops.fused_rollout import make_counter
+
+        ext = ops._load_extension()
+        buf = algo.replay_buffer
+        buf._sync_size_dev()
+        dev = buf.device
+        st = buf._storage
+        cap = buf.buffer_size
+        act_flat = st["actions"].view(cap, -1)
+        twin = hasattr(algo, "q_function_1")
+        delay = getattr(algo, "policy_delay", 1)
+        gamma = float(algo.gamma)
+        rho = float(algo.polyak_rho)
+        seed = _stream_seed()
+        self.ctr = make_counter(dev)
+        self.num_iters = num_iters
+        self.twin = twin
+
+        import numpy as np
+
+        action_limit = float(np.asarray(algo.env.action_space.high).reshape(-1)[0])
+
+        if twin:
+            q_fns = [algo.q_function_1, algo.q_function_2]
+            tq_fns = [algo.target_q_function_1, algo.target_q_function_2]
+        else:
+            q_fns = [algo.q_function]
+            tq_fns = [algo.target_q_function]
+
+        n_policy = (num_iters + delay - 1) // delay
+        self.q_losses = [torch.zeros(num_iters, device=dev) for _ in q_fns]
+        self.pi_losses = torch.zeros(n_policy, device=dev)
+        self.all_q = [
+            torch.zeros(num_iters, minibatch_size, device=dev) for _ in q_fns
+        ]
+
+        polyak_src = list(algo.policy.network.parameters())
+        polyak_dst = list(algo.target_policy.network.parameters())
+        for q, tq in zip(q_fns, tq_fns):
+            polyak_src += list(q.network.parameters())
+            polyak_dst += list(tq.network.parameters())
+        polyak_src = [p.data for p in polyak_src]
+        polyak_dst = [p.data for p in polyak_dst]
+
+        def body():
+            pi_k = 0
+            for i in range(num_iters):
+                qin, obs, nxt, rew, dn = ext.replay_gather(
+                    st["observations"], act_flat, st["rewards"],
+                    st["next_observations"], st["dones"], buf._size_dev,
+                    minibatch_size, seed, 3 * i, self.ctr,
+                )
+                # pre-update Q values for the epoch metrics
+                # (reference td3.py:230-236)
+                for q, buf_q in zip(q_fns, self.all_q):
+                    buf_q[i].copy_(forward_only(q, qin).view(-1))
+                # target chain
+                na = forward_only(algo.target_policy, nxt)
+                if twin:
+                    na = ext.td3_smooth(
+                        na, seed, 3 * i + 1,
+                        float(algo.target_noise_scale),
+                        float(algo.target_noise_clip), action_limit, self.ctr,
+                    )
+                qt_in = torch.cat([nxt, na], dim=-1).contiguous()
+                if twin:
+                    q1t = forward_only(tq_fns[0], qt_in).view(-1)
+                    q2t = forward_only(tq_fns[1], qt_in).view(-1)
+                    targets = ext.q_target_min2(rew, dn, q1t, q2t, gamma)
+                else:
+                    qt = forward_only(tq_fns[0], qt_in).view(-1)
+                    targets = ext.q_target(rew, dn, qt, gamma)
+                for q, buf_l in zip(q_fns, self.q_losses):
+                    buf_l[i].copy_(
+                        q_step(q, obs, None, targets, _noop_hook, qin=qin,
+                               step_delta=float(i))
+                    )
+                if i % delay == 0:
+                    self.pi_losses[pi_k].copy_(
+                        policy_step(algo.policy, q_fns[0], obs, _noop_hook,
+                                    step_delta=float(pi_k))
+                    )
+                    ext.fused_polyak_(polyak_src, polyak_dst, rho)
+                    pi_k += 1
+            for q in q_fns:
+                q.optimizer.bump_steps(float(num_iters))
+            algo.policy.optimizer.bump_steps(float(pi_k))
+            ext.counter_add_(self.ctr, 3 * num_iters)
+
+        state = [t for m in ([algo.policy, algo.target_policy] + q_fns + tq_fns)
+                 for t in (p.data for p in m.network.parameters())]
+        state += _ensure_adam_state(algo.policy.optimizer)
+        for q in q_fns:
+            state += _ensure_adam_state(q.optimizer)
+        state.append(self.ctr)
+        self.loop = _CapturedLoop(body, state)
+
+    def run(self) -> Dict[str, float]:
+        self.loop.replay()
+        metrics: Dict[str, float] = {
+            "policy/average_loss": float(self.pi_losses.mean()),
+        }
+        names = ["q-function_1", "q-function_2"] if self.twin else ["q-function"]
+        for name, losses, qv in zip(names, self.q_losses, self.all_q):
+            metrics[f"{name}/average_loss"] = float(losses.mean())
+            metrics[f"{name}/avarage_q-value"] = float(qv.mean())
+            metrics[f"{name}/max_q-value"] = float(qv.max())
+            metrics[f"{name}/min_q-value"] = float(qv.min())
+        return metrics
+
+
+def graphed_epoch(algo, num_train_steps: int, minibatch_size: int) -> Dict[str, float]:
+    """Run (capturing on first use) the graphed off-policy epoch."""
+    from rl_replicas_amd.ops.fused_onpolicy import _get_cached_graph
+
+    key = (
+        num_train_steps,
+        minibatch_size,
+        id(algo.replay_buffer),
+        tuple(id(p) for p in algo.policy.network.parameters()),
+        ops.compute_bf16(),
+    )
+    graphed = _get_cached_graph(
+        algo, "_offpolicy_graph", key,
+        lambda: _GraphedOffPolicy(algo, num_train_steps, minibatch_size),
+    )
+    return graphed.run()

@@ -91,6 +91,14 @@ __global__ void segmented_gae_kernel(const float* rewards, const float* values,
 __global__ void normalize_kernel(const float* x, float* y, int n);
 __global__ void q_target_kernel(const float* r, const float* d, const float* qn,
                                 float* out, float gamma, int n);
+__global__ void td3_smooth_kernel(const float* a, float* out, int total,
+                                  uint64_t seed, uint64_t offset, float scale,
+                                  float clip, float limit,
+                                  const unsigned long long* offset_ptr);
+__global__ void q_target_min2_kernel(const float* r, const float* d,
+                                     const float* q1, const float* q2,
+                                     float* out, float gamma, int n);
+__global__ void replay_gather_kernel(ReplayGatherArgs a);
 
 __global__ void fused_adam_kernel(AdamArgs a);
 __global__ void adam_step_bump_kernel(AdamArgs a);
@@ -852,6 +860,79 @@ torch::Tensor synthetic_env_reset(int64_t num_envs, int64_t obs_dim,
   return s_out;
 }
 
+torch::Tensor td3_smooth(torch::Tensor a, int64_t seed, int64_t offset,
+                         double scale, double clip, double limit,
+                         c10::optional<torch::Tensor> offset_ctr) {
+  check_f32_gpu(a, "actions");
+  auto out = torch::empty_like(a);
+  const int total = (int)a.numel();
+  hipLaunchKernelGGL(td3_smooth_kernel,
+                     dim3(std::min(256, (total + 255) / 256)), dim3(256), 0,
+                     current_stream(), a.data_ptr<float>(),
+                     out.data_ptr<float>(), total, (uint64_t)seed,
+                     (uint64_t)offset, (float)scale, (float)clip,
+                     (float)limit, ctr_ptr(offset_ctr));
+  HIP_OK(hipGetLastError());
+  return out;
+}
+
+torch::Tensor q_target_min2(torch::Tensor r, torch::Tensor d, torch::Tensor q1,
+                            torch::Tensor q2, double gamma) {
+  check_f32_gpu(r, "rewards");
+  auto out = torch::empty_like(r);
+  const int n = (int)r.numel();
+  hipLaunchKernelGGL(q_target_min2_kernel, dim3((n + 255) / 256), dim3(256), 0,
+                     current_stream(), r.data_ptr<float>(), d.data_ptr<float>(),
+                     q1.data_ptr<float>(), q2.data_ptr<float>(),
+                     out.data_ptr<float>(), (float)gamma, n);
+  HIP_OK(hipGetLastError());
+  return out;
+}
+
+std::vector<torch::Tensor> replay_gather(torch::Tensor obs, torch::Tensor act,
+                                         torch::Tensor rew, torch::Tensor nxt,
+                                         torch::Tensor dn, torch::Tensor size_dev,
+                                         int64_t B, int64_t seed, int64_t offset,
+                                         c10::optional<torch::Tensor> offset_ctr) {
+  check_f32_gpu(obs, "obs");
+  check_f32_gpu(act, "act");
+  check_f32_gpu(rew, "rew");
+  check_f32_gpu(nxt, "nxt");
+  check_f32_gpu(dn, "dn");
+  TORCH_CHECK(size_dev.scalar_type() == torch::kInt64 && size_dev.is_cuda() &&
+                  size_dev.numel() == 1,
+              "size_dev must be a 1-element int64 CUDA tensor");
+  ReplayGatherArgs a{};
+  a.obs = obs.data_ptr<float>();
+  a.act = act.data_ptr<float>();
+  a.rew = rew.data_ptr<float>();
+  a.nxt = nxt.data_ptr<float>();
+  a.dn = dn.data_ptr<float>();
+  a.size = reinterpret_cast<const long long*>(size_dev.data_ptr<int64_t>());
+  a.B = (int)B;
+  a.O = (int)obs.size(1);
+  a.A = (int)act.size(1);
+  auto qin = torch::empty({B, a.O + a.A}, obs.options());
+  auto obs_out = torch::empty({B, a.O}, obs.options());
+  auto nxt_out = torch::empty({B, a.O}, obs.options());
+  auto rew_out = torch::empty({B}, obs.options());
+  auto dn_out = torch::empty({B}, obs.options());
+  a.qin = qin.data_ptr<float>();
+  a.obs_out = obs_out.data_ptr<float>();
+  a.nxt_out = nxt_out.data_ptr<float>();
+  a.rew_out = rew_out.data_ptr<float>();
+  a.dn_out = dn_out.data_ptr<float>();
+  a.seed = (uint64_t)seed;
+  a.offset = (uint64_t)offset;
+  a.offset_ptr = ctr_ptr(offset_ctr);
+  const long total = (long)a.B * (2 * a.O + a.A + 2);
+  hipLaunchKernelGGL(replay_gather_kernel,
+                     dim3((int)std::min<long>(512, (total + 255) / 256)),
+                     dim3(256), 0, current_stream(), a);
+  HIP_OK(hipGetLastError());
+  return {qin, obs_out, nxt_out, rew_out, dn_out};
+}
+
 void counter_add_(torch::Tensor ctr, int64_t delta) {
   TORCH_CHECK(ctr.scalar_type() == torch::kInt64 && ctr.is_cuda() && ctr.numel() == 1,
               "ctr must be a 1-element int64 CUDA tensor");
@@ -883,6 +964,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");
   m.def("normalize", &normalize, "fused mean/std normalize (gfx950)");
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");
+  m.def("td3_smooth", &td3_smooth,
+        "TD3 target-policy smoothing: Philox noise + clip + limit (gfx950)",
+        py::arg("a"), py::arg("seed"), py::arg("offset"), py::arg("scale"),
+        py::arg("clip"), py::arg("limit"), py::arg("offset_ctr") = py::none());
+  m.def("q_target_min2", &q_target_min2,
+        "fused min-twin Q-learning target (gfx950)");
+  m.def("replay_gather", &replay_gather,
+        "one-kernel Philox minibatch gather from the HBM replay ring (gfx950)",
+        py::arg("obs"), py::arg("act"), py::arg("rew"), py::arg("nxt"),
+        py::arg("dn"), py::arg("size_dev"), py::arg("B"), py::arg("seed"),
+        py::arg("offset"), py::arg("offset_ctr") = py::none());
   m.def("fused_adam_", &fused_adam_, "fused multi-tensor Adam (gfx950)",
         py::arg("params"), py::arg("grads"), py::arg("exp_avgs"),
         py::arg("exp_avg_sqs"), py::arg("steps"), py::arg("lr"),

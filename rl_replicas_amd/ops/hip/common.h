@@ -113,3 +113,21 @@ struct PolyakArgs {
   int n_tensors;
   float rho;
 };
+
+// replay-ring minibatch gather (offpolicy_kernels.hip / bindings.hip)
+struct ReplayGatherArgs {
+  const float* obs;   // [cap, O] ring storage
+  const float* act;   // [cap, A]
+  const float* rew;   // [cap]
+  const float* nxt;   // [cap, O]
+  const float* dn;    // [cap]
+  const long long* size;  // device scalar: current ring fill
+  float* qin;      // [B, O+A]  critic input (obs|act)
+  float* obs_out;  // [B, O]    actor input
+  float* nxt_out;  // [B, O]
+  float* rew_out;  // [B]
+  float* dn_out;   // [B]
+  int B, O, A;
+  uint64_t seed, offset;
+  const unsigned long long* offset_ptr;  // optional device RNG counter
+};

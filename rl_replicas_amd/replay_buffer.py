@@ -33,6 +33,16 @@ class ReplayBuffer:
         self.current_size = 0
         self._write = 0
         self._storage: Optional[Dict[str, torch.Tensor]] = None
+        # device mirror of current_size, read by the fused gather kernel
+        # (lets one captured hipGraph keep sampling as the ring fills)
+        self._size_dev: Optional[torch.Tensor] = None
+        self._gather_offset = 0
+
+    def _sync_size_dev(self) -> None:
+        if self.device is not None and self.device.type == "cuda":
+            if self._size_dev is None:
+                self._size_dev = torch.zeros(1, dtype=torch.int64, device=self.device)
+            self._size_dev.fill_(self.current_size)
 
     # ------------------------------------------------------------------
     def _allocate(self, obs_dim: int, act_shape) -> None:
@@ -98,6 +108,7 @@ class ReplayBuffer:
                 self._storage[key][: n - first] = t[first:]
         self._write = (pos + n) % self.buffer_size
         self.current_size = min(self.current_size + n, self.buffer_size)
+        self._sync_size_dev()
 
     def _add_flat_tensors(self, flat: Dict[str, torch.Tensor]) -> None:
         batch = {
@@ -124,6 +135,7 @@ class ReplayBuffer:
                 self._storage[key][: n - first] = t[first:]
         self._write = (pos + n) % self.buffer_size
         self.current_size = min(self.current_size + n, self.buffer_size)
+        self._sync_size_dev()
 
     # ------------------------------------------------------------------
     def sample_minibatch_tensors(self, minibatch_size: int = 32) -> Dict[str, torch.Tensor]:
@@ -141,6 +153,50 @@ class ReplayBuffer:
         """Reference-compatible numpy minibatch (reference :51-74)."""
         mb = self.sample_minibatch_tensors(minibatch_size)
         return {k: v.cpu().numpy() for k, v in mb.items()}
+
+    # ------------------------------------------------------------------
+    def gather_minibatch_fused(
+        self,
+        minibatch_size: int,
+        seed: int,
+        offset: int,
+        offset_ctr: Optional[torch.Tensor] = None,
+    ) -> Dict[str, torch.Tensor]:
+        """ONE-kernel minibatch draw+gather from the HBM ring (GPU only).
+
+        Philox-indexed (stateless given seed/offset, graph-replay safe
+        via `offset_ctr`); returns the critic input `qin` = [obs|act]
+        pre-concatenated alongside the standard keys — replaces
+        randint + 5 index_selects + torch.cat on the DDPG/TD3 hot path
+        (reference ddpg.py:222-228)."""
+        from rl_replicas_amd import ops
+
+        assert self._storage is not None and self.current_size > 0, "empty replay buffer"
+        assert self.device is not None and self.device.type == "cuda"
+        self._sync_size_dev()
+        ext = ops._load_extension()
+        st = self._storage
+        cap = self.buffer_size
+        qin, obs, nxt, rew, dn = ext.replay_gather(
+            st["observations"],
+            st["actions"].view(cap, -1),
+            st["rewards"],
+            st["next_observations"],
+            st["dones"],
+            self._size_dev,
+            int(minibatch_size),
+            int(seed),
+            int(offset),
+            offset_ctr,
+        )
+        return {
+            "qin": qin,
+            "observations": obs,
+            "actions": qin[:, st["observations"].shape[1] :],
+            "rewards": rew,
+            "next_observations": nxt,
+            "dones": dn,
+        }
 
     def __len__(self) -> int:
         return self.current_size

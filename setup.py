@@ -27,6 +27,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "env_kernels.hip"),
         os.path.join(HIP_DIR, "rollout_kernels.hip"),
         os.path.join(HIP_DIR, "update_kernels.hip"),
+        os.path.join(HIP_DIR, "offpolicy_kernels.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
