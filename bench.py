@@ -92,9 +92,56 @@ def build_model(device: str, num_envs: int, seed: int, env_mode: str = "device")
     return model, sampler
 
 
+def build_model_td3(device: str, num_envs: int, seed: int):
+    """BASELINE config #3: TD3 Ant-v4 on one MI355X — the reference
+    run_td3.py config (policy [obs,256,256,act] ReLU+Tanh, twin Q
+    [obs+act,256,256,1] ReLU, Adam 1e-3, buffer 1e6, 50 env steps + 50
+    train iterations of minibatch 100 per epoch)."""
+    from rl_replicas_amd import envs, ops
+    from rl_replicas_amd.algorithms import TD3
+    from rl_replicas_amd.evaluator import Evaluator
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.policies import DeterministicPolicy, RandomPolicy
+    from rl_replicas_amd.q_function import QFunction
+    from rl_replicas_amd.replay_buffer import ReplayBuffer
+    from rl_replicas_amd.samplers import DeviceSampler
+
+    obs_dim, act_dim = envs.MUJOCO_SHAPES["Ant-v4"]
+    pnet = MLP(
+        [obs_dim, 256, 256, act_dim],
+        activation_function=nn.ReLU,
+        output_activation_function=nn.Tanh,
+    ).to(device)
+    policy = DeterministicPolicy(pnet, ops.make_adam(pnet.parameters(), lr=1e-3))
+    qs = []
+    for _ in range(2):
+        qn = MLP([obs_dim + act_dim, 256, 256, 1], activation_function=nn.ReLU).to(device)
+        qs.append(QFunction(qn, ops.make_adam(qn.parameters(), lr=1e-3)))
+    venv = envs.DeviceVectorEnv("Ant-v4", num_envs=num_envs, device=device)
+    sampler = DeviceSampler(venv, seed=seed, is_continuous=True)
+    model = TD3(
+        policy,
+        RandomPolicy(venv.action_space),
+        qs[0],
+        qs[1],
+        venv,
+        sampler,
+        ReplayBuffer(int(1e6), device=device),
+        Evaluator(seed=seed),
+    )
+    return model, sampler
+
+
 def main() -> None:
     parser = argparse.ArgumentParser()
     parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument(
+        "--algo",
+        choices=["ppo", "td3"],
+        default="ppo",
+        help="flagship PPO HalfCheetah (driver contract) or TD3 Ant-v4 "
+        "(BASELINE config #3; step = 50 env steps + 50 train iterations)",
+    )
     parser.add_argument("--steps", type=int, default=None, help="timed PPO epochs (default: 300 on GPU so the timed region is seconds long, 3 on CPU)")
     parser.add_argument("--warmup", type=int, default=None, help="untimed warmup epochs (default: 20 on GPU, 1 on CPU)")
     parser.add_argument("--batch-per-gpu", type=int, default=4000)
@@ -149,7 +196,15 @@ def main() -> None:
     torch.manual_seed(1234 + rank)
     np.random.seed(1234 + rank)
 
-    model, sampler = build_model(device, args.num_envs, seed=1234 + rank, env_mode=args.env)
+    if args.algo == "td3":
+        # reference off-policy protocol: 50 env steps + 50 train
+        # iterations (minibatch 100) per epoch (run_td3.py, td3.py:94-105)
+        td3_batch = 50
+        if args.num_envs > td3_batch:
+            args.num_envs = td3_batch
+        model, sampler = build_model_td3(device, args.num_envs, seed=1234 + rank)
+    else:
+        model, sampler = build_model(device, args.num_envs, seed=1234 + rank, env_mode=args.env)
     if world > 1:
         enable_data_parallel(model)
 
@@ -161,7 +216,7 @@ def main() -> None:
     phase_ms = {"sample": 0.0, "train": 0.0}
     epoch_returns = []  # per-epoch mean episode return (learning evidence)
 
-    def one_epoch():
+    def one_epoch_ppo():
         t0 = time.perf_counter()
         experience = sampler.sample(args.batch_per_gpu, model.policy)
         if use_gpu and args.phase_timing:
@@ -175,6 +230,26 @@ def main() -> None:
         phase_ms["train"] += (time.perf_counter() - t1) * 1000.0
         if experience.episode_returns:
             epoch_returns.append(float(np.mean(experience.episode_returns)))
+
+    def one_epoch_td3():
+        t0 = time.perf_counter()
+        experience = sampler.sample(50, model.noised_policy)
+        model.replay_buffer.add_experience(experience)
+        if use_gpu and args.phase_timing:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        model.current_total_steps += sum(experience.episode_lengths)
+        model.train(model.replay_buffer, num_train_steps=50, minibatch_size=100)
+        if use_gpu and args.phase_timing:
+            torch.cuda.synchronize()
+        phase_ms["sample"] += (t1 - t0) * 1000.0
+        phase_ms["train"] += (time.perf_counter() - t1) * 1000.0
+        if experience.episode_returns:
+            epoch_returns.append(float(np.mean(experience.episode_returns)))
+
+    one_epoch = one_epoch_td3 if args.algo == "td3" else one_epoch_ppo
+    if args.algo == "td3":
+        args.batch_per_gpu = 50  # env steps per epoch, reference protocol
 
     def barrier_sync():
         if world > 1:
@@ -239,17 +314,32 @@ def main() -> None:
                 "are impossible offline (BASELINE.md); algorithm math is instead "
                 "pinned to the reference by tests/test_reference_equivalence.py.",
             },
-            "config": {
-                "model": "PPO HalfCheetah-v4 (policy MLP [17,64,32,6] tanh, value [17,64,32,1])",
-                "global_batch": world * args.batch_per_gpu,
-                "seq_len": 1000,
-                "parallelism": f"dp{world}",
-                "num_envs_per_gpu": args.num_envs,
-                "env_residency": args.env,
-                "policy_grads_per_epoch": 80,
-                "value_grads_per_epoch": 80,
-                "note": "step = one PPO epoch (sample batch + full update); vs_baseline is the reference's implied ~946 env-steps/s serial CPU throughput (BASELINE.md)",
-            },
+            "config": (
+                {
+                    "model": "PPO HalfCheetah-v4 (policy MLP [17,64,32,6] tanh, value [17,64,32,1])",
+                    "global_batch": world * args.batch_per_gpu,
+                    "seq_len": 1000,
+                    "parallelism": f"dp{world}",
+                    "num_envs_per_gpu": args.num_envs,
+                    "env_residency": args.env,
+                    "policy_grads_per_epoch": 80,
+                    "value_grads_per_epoch": 80,
+                    "note": "step = one PPO epoch (sample batch + full update); vs_baseline is the reference's implied ~946 env-steps/s serial CPU throughput (BASELINE.md)",
+                }
+                if args.algo == "ppo"
+                else {
+                    "model": "TD3 Ant-v4 (policy MLP [27,256,256,8] ReLU+Tanh, twin Q [35,256,256,1] ReLU)",
+                    "global_batch": world * args.batch_per_gpu,
+                    "seq_len": 1000,
+                    "parallelism": f"dp{world}",
+                    "num_envs_per_gpu": args.num_envs,
+                    "env_residency": "device",
+                    "train_iterations_per_epoch": 50,
+                    "minibatch_size": 100,
+                    "replay_buffer": 1_000_000,
+                    "note": "step = one TD3 epoch (50 env steps + 50 twin-critic train iterations, reference run_td3.py protocol); no published reference throughput for TD3 (BASELINE.md) — vs_baseline uses the same 946 serial-CPU figure as a floor",
+                }
+            ),
         }
         print(json.dumps(result))
 
