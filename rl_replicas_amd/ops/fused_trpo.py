@@ -59,10 +59,112 @@ def _mlp_jvp(obs: Tensor, weights, hidden, final_out, acts,
     return t
 
 
+class CapturableFVP:
+    """Analytic v -> (H + damping I) v with CAPTURE-STABLE buffers.
+
+    The saved activations (and inv_var / probs) live in tensors owned by
+    this object; `refresh(obs)` re-runs the fused forward and copies the
+    results in.  `__call__` is then a fixed sequence of device kernels
+    reading those buffers — so the whole CG solve that calls it can be
+    captured ONCE into a hipGraph and replayed every epoch
+    (ConjugateGradientOptimizer._CapturedCG), with only the buffer
+    refresh running eagerly per epoch.
+    """
+
+    def __init__(self, policy, kind: str, obs: Tensor, damping: float,
+                 weights, biases, acts, slots: List[tuple], log_std):
+        self.kind = kind
+        self.damping = damping
+        self.weights = weights
+        self.biases = biases
+        self.acts = acts
+        self.slots = slots
+        self.log_std = log_std
+        self.B = obs.shape[0]
+        self.obs = torch.empty_like(obs)
+        ext = ops._load_extension()
+        outs = ext.mlp_forward(obs.contiguous(), list(weights), list(biases),
+                               acts, True, ops.compute_bf16())
+        self.final_out = torch.empty_like(outs[0])
+        self.hidden = [torch.empty_like(h) for h in outs[1:]]
+        if kind == "gaussian":
+            self.inv_var = torch.empty_like(log_std.detach())
+        else:
+            self.probs = torch.empty_like(outs[0])
+        self.numels: List[int] = []
+        for s in slots:
+            if s[0] == "w":
+                self.numels.append(weights[s[1]].numel())
+            elif s[0] == "b":
+                self.numels.append(biases[s[1]].numel())
+            else:
+                self.numels.append(log_std.numel())
+        self.graph_key = (
+            kind, tuple(obs.shape), float(damping),
+            tuple(id(w) for w in weights) + tuple(id(b) for b in biases),
+            ops.compute_bf16(),
+        )
+        self.refresh(obs)
+
+    @torch.no_grad()
+    def refresh(self, obs: Tensor) -> "CapturableFVP":
+        ext = ops._load_extension()
+        obs = obs.contiguous()
+        self.obs.copy_(obs)
+        outs = ext.mlp_forward(self.obs, list(self.weights), list(self.biases),
+                               self.acts, True, ops.compute_bf16())
+        self.final_out.copy_(outs[0])
+        for buf, h in zip(self.hidden, outs[1:]):
+            buf.copy_(h)
+        if self.kind == "gaussian":
+            self.inv_var.copy_(torch.exp(-2.0 * self.log_std.detach()))
+        else:
+            self.probs.copy_(torch.softmax(self.final_out, dim=-1))
+        return self
+
+    def __call__(self, v: Tensor) -> Tensor:
+        ext = ops._load_extension()
+        weights, biases = self.weights, self.biases
+        parts = torch.split(v, self.numels)
+        v_ws: List[Optional[Tensor]] = [None] * len(weights)
+        v_bs: List[Optional[Tensor]] = [None] * len(biases)
+        v_ls: Optional[Tensor] = None
+        for s, chunk in zip(self.slots, parts):
+            if s[0] == "w":
+                v_ws[s[1]] = chunk.view(weights[s[1]].shape)
+            elif s[0] == "b":
+                v_bs[s[1]] = chunk.view(biases[s[1]].shape)
+            else:
+                v_ls = chunk
+
+        t = _mlp_jvp(self.obs, weights, self.hidden, self.final_out, self.acts,
+                     v_ws, v_bs)
+        if self.kind == "gaussian":
+            u = t * self.inv_var / self.B
+        else:
+            u = (t * self.probs
+                 - self.probs * (t * self.probs).sum(dim=-1, keepdim=True)) / self.B
+        grads = ext.mlp_backward(u.contiguous(), self.obs, list(weights),
+                                 list(biases), self.hidden, self.final_out,
+                                 self.acts, ops.compute_bf16())
+        n = len(weights)
+        dws = grads[1 : 1 + n]
+        dbs = grads[1 + n :]
+        out_parts = []
+        for s in self.slots:
+            if s[0] == "w":
+                out_parts.append(dws[s[1]].reshape(-1))
+            elif s[0] == "b":
+                out_parts.append(dbs[s[1]].reshape(-1))
+            else:
+                out_parts.append(2.0 * v_ls)
+        return torch.cat(out_parts) + self.damping * v
+
+
 def make_fvp(policy, obs: Tensor, damping: float) -> Optional[Callable[[Tensor], Tensor]]:
-    """Build v -> (H + damping I) v for a Gaussian/Categorical MLP policy,
-    or None if the policy/param layout isn't supported (caller falls
-    back to the double-backward FVP)."""
+    """Build (or refresh the policy-cached) analytic FVP for a
+    Gaussian/Categorical MLP policy; None if the policy/param layout
+    isn't supported (caller falls back to the double-backward FVP)."""
     from rl_replicas_amd.ops import fused_onpolicy as fop
 
     if not (obs.is_cuda and ops.hip_available()):
@@ -99,58 +201,13 @@ def make_fvp(policy, obs: Tensor, damping: float) -> Optional[Callable[[Tensor],
     if kind == "gaussian" and not any(s[0] == "log_std" for s in slots):
         return None
 
-    ext = ops._load_extension()
-    obs = obs.contiguous()
-    B = obs.shape[0]
-    outs = ext.mlp_forward(obs, list(weights), list(biases), acts, True,
-                           ops.compute_bf16())
-    final_out, hidden = outs[0], list(outs[1:])
-
-    if kind == "gaussian":
-        inv_var = torch.exp(-2.0 * log_std.detach())  # [D]
-    else:
-        probs = torch.softmax(final_out, dim=-1)  # [B, N]
-
-    numels: List[int] = []
-    for s in slots:
-        if s[0] == "w":
-            numels.append(weights[s[1]].numel())
-        elif s[0] == "b":
-            numels.append(biases[s[1]].numel())
-        else:
-            numels.append(log_std.numel())
-
-    def fvp(v: Tensor) -> Tensor:
-        parts = torch.split(v, numels)
-        v_ws: List[Optional[Tensor]] = [None] * len(weights)
-        v_bs: List[Optional[Tensor]] = [None] * len(biases)
-        v_ls: Optional[Tensor] = None
-        for s, chunk in zip(slots, parts):
-            if s[0] == "w":
-                v_ws[s[1]] = chunk.view(weights[s[1]].shape)
-            elif s[0] == "b":
-                v_bs[s[1]] = chunk.view(biases[s[1]].shape)
-            else:
-                v_ls = chunk
-
-        t = _mlp_jvp(obs, weights, hidden, final_out, acts, v_ws, v_bs)
-        if kind == "gaussian":
-            u = t * inv_var / B
-        else:
-            u = (t * probs - probs * (t * probs).sum(dim=-1, keepdim=True)) / B
-        grads = ext.mlp_backward(u.contiguous(), obs, list(weights), list(biases),
-                                 hidden, final_out, acts, ops.compute_bf16())
-        n = len(weights)
-        dws = grads[1 : 1 + n]
-        dbs = grads[1 + n :]
-        out_parts = []
-        for s in slots:
-            if s[0] == "w":
-                out_parts.append(dws[s[1]].reshape(-1))
-            elif s[0] == "b":
-                out_parts.append(dbs[s[1]].reshape(-1))
-            else:
-                out_parts.append(2.0 * v_ls)
-        return torch.cat(out_parts) + damping * v
-
+    cached = getattr(policy, "_fvp_cache", None)
+    key = (kind, tuple(obs.shape), float(damping),
+           tuple(id(w) for w in weights) + tuple(id(b) for b in biases),
+           ops.compute_bf16())
+    if cached is not None and cached.graph_key == key and cached.slots == slots:
+        return cached.refresh(obs)
+    fvp = CapturableFVP(policy, kind, obs, damping, weights, biases, acts,
+                        slots, log_std)
+    policy._fvp_cache = fvp
     return fvp

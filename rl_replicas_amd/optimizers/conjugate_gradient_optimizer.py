@@ -47,6 +47,33 @@ def _flatten(tensors: Iterable[Tensor]) -> Tensor:
     return torch.cat([t.reshape(-1) for t in tensors])
 
 
+class _CapturedCG:
+    """The masked fixed-iteration CG solve + NaN guard + quadratic form
+    captured as ONE hipGraph (inputs: the capture-stable `b` buffer and
+    the FVP's activation buffers; outputs: direction, dHd+1e-8)."""
+
+    def __init__(self, opt: "ConjugateGradientOptimizer", hvp, loss_grad: Tensor):
+        from rl_replicas_amd.ops.fused_onpolicy import _CapturedLoop
+
+        self.hvp = hvp
+        self.b = loss_grad.detach().clone()
+        outer = self
+
+        def body():
+            x = opt._conjugate_gradient(outer.hvp, outer.b)
+            direction = torch.nan_to_num(x, nan=0.0)
+            quad = torch.dot(direction, outer.hvp(direction)) + 1e-8
+            return direction, quad
+
+        # body is pure (reads buffers, writes fresh outputs): no state
+        # to snapshot/restore around the warmup runs
+        self.loop = _CapturedLoop(body, [])
+
+    def run(self, loss_grad: Tensor):
+        self.b.copy_(loss_grad)
+        return self.loop.replay()
+
+
 class ConjugateGradientOptimizer(Optimizer):
     """Constrained step: x = H^-1 g via CG, scaled to the KL trust region.
 
@@ -109,12 +136,20 @@ class ConjugateGradientOptimizer(Optimizer):
             hvp = lambda v: reduce_hook(local_hvp(v))  # noqa: E731
             loss_function = lambda: reduce_hook(local_loss())  # noqa: E731
             kl_divergence_function = lambda: reduce_hook(local_kl())  # noqa: E731
-        direction = self._conjugate_gradient(hvp, loss_grad)
-        # NaN guard on the direction (reference :83)
-        direction = torch.nan_to_num(direction, nan=0.0)
 
-        # beta = sqrt(2*delta / (d^T H d)) (reference :86-90)
-        quad = torch.dot(direction, hvp(direction)) + 1e-8
+        solver = None
+        if reduce_hook is None and loss_grad.is_cuda:
+            solver = self._get_captured_cg(hvp, loss_grad)
+        if solver is not None:
+            # whole CG solve (+ nan guard + quadratic form) = ONE hipGraph
+            # replay; the sqrt readback below is the solve's single sync
+            direction, quad = solver.run(loss_grad)
+        else:
+            direction = self._conjugate_gradient(hvp, loss_grad)
+            # NaN guard on the direction (reference :83)
+            direction = torch.nan_to_num(direction, nan=0.0)
+            # beta = sqrt(2*delta / (d^T H d)) (reference :86-90)
+            quad = torch.dot(direction, hvp(direction)) + 1e-8
         step_size = float(torch.sqrt(2.0 * self.max_constraint / quad))
         if step_size != step_size:  # NaN guard (reference :92-94)
             step_size = 1.0
@@ -122,6 +157,29 @@ class ConjugateGradientOptimizer(Optimizer):
         self._backtracking_line_search(
             params, step_size * direction, loss_function, kl_divergence_function
         )
+
+    # ------------------------------------------------------------------
+    def _get_captured_cg(self, hvp, loss_grad: Tensor):
+        """hipGraph-captured CG solve, cached across epochs.
+
+        Only for an FVP with capture-stable buffers (fused_trpo.
+        CapturableFVP exposes `graph_key`); the double-backward FVP
+        builds autograd graphs and cannot be captured."""
+        import os
+
+        key = getattr(hvp, "graph_key", None)
+        if key is None or os.environ.get("RL_REPLICAS_AMD_DISABLE_GRAPHS", "0") == "1":
+            return None
+        full_key = (key, int(loss_grad.numel()), self.n_conjugate_gradients)
+        cached = getattr(self, "_cg_graph", None)
+        # identity check matters: the captured graph bakes the FVP's
+        # buffer POINTERS — a rebuilt FVP object (new buffers) needs a
+        # fresh capture even if its key matches
+        if cached is not None and cached[0] == full_key and cached[1].hvp is hvp:
+            return cached[1]
+        solver = _CapturedCG(self, hvp, loss_grad)
+        self._cg_graph = (full_key, solver)
+        return solver
 
     # ------------------------------------------------------------------
     def _make_fisher_vector_product(
@@ -150,25 +208,38 @@ class ConjugateGradientOptimizer(Optimizer):
     def _conjugate_gradient(
         self, hvp: Callable[[Tensor], Tensor], b: Tensor, residual_tol: float = 1e-10
     ) -> Tensor:
-        """Solve H x = b (standard CG; reference :169-202).
+        """Solve H x = b (standard CG; reference :169-202) with ZERO
+        host synchronization inside the loop.
 
-        All vectors stay on b's device; the host loop only reads the
-        scalar residual for the early-exit test.
+        The reference's early exit (`if r_dot_r < tol: break`) forces a
+        device->host readback per iteration; here the loop runs a FIXED
+        n_conjugate_gradients iterations and the exit is a device-side
+        mask: once the residual drops below tol (or goes NaN, which the
+        reference also never recovers from), every later update
+        multiplies by zero / keeps the frozen state, so the returned x
+        is the one the reference would have returned — without a single
+        `.item()`-class sync.  This also makes the whole solve
+        hipGraph-capturable (fixed kernel sequence).
         """
         x = torch.zeros_like(b)
         r = b.clone()
         p = b.clone()
         r_dot_r = torch.dot(r, r)
+        one = torch.ones((), device=b.device, dtype=b.dtype)
+        zero = torch.zeros((), device=b.device, dtype=b.dtype)
+        # iteration 0 always runs (the reference checks AFTER the update)
+        active = torch.ones((), device=b.device, dtype=torch.bool)
         for _ in range(self.n_conjugate_gradients):
             hp = hvp(p)
-            alpha = r_dot_r / torch.dot(p, hp)
+            p_dot_hp = torch.dot(p, hp)
+            alpha = torch.where(active, r_dot_r / p_dot_hp, zero)
             x = x + alpha * p
             r = r - alpha * hp
             new_r_dot_r = torch.dot(r, r)
-            p = r + (new_r_dot_r / r_dot_r) * p
-            r_dot_r = new_r_dot_r
-            if float(r_dot_r) < residual_tol:
-                break
+            beta = new_r_dot_r / torch.where(r_dot_r == 0, one, r_dot_r)
+            p = torch.where(active, r + beta * p, p)
+            r_dot_r = torch.where(active, new_r_dot_r, r_dot_r)
+            active = active & (r_dot_r >= residual_tol)
         return x
 
     # ------------------------------------------------------------------
