@@ -37,9 +37,9 @@ void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
                          hipStream_t stream);
 void launch_gaussian_loss(const float* mean, const float* actions,
                           const float* old_logp, const float* adv,
-                          const float* log_std, float* dmean, float* partials,
-                          int B, int D, float clip, int mode, int n_blocks,
-                          hipStream_t stream);
+                          const float* log_std, float* dmean, float* c_buf,
+                          float* partials, int B, int D, float clip, int mode,
+                          int n_blocks, hipStream_t stream);
 __global__ void loss_partials_finalize(const float* partials, float* dlog_std,
                                        float* scalars, int n_blocks, int D);
 __global__ void gaussian_logp_kernel(const float* mean, const float* actions,
@@ -410,18 +410,23 @@ std::vector<torch::Tensor> gaussian_policy_loss(torch::Tensor mean,
   check_f32_gpu(mean, "mean");
   const int B = (int)mean.size(0);
   const int D = (int)mean.size(1);
-  TORCH_CHECK(D >= 1 && D <= 8, "action dim must be in [1,8] for the fused loss");
+  TORCH_CHECK(D >= 1 && D <= 512,
+              "action dim must be in [1,512] for the fused loss");
   auto opts = mean.options();
   auto dmean = torch::empty_like(mean);
   auto dlog_std = torch::empty({D}, opts);
   auto scalars = torch::empty({1}, opts);
   const int n_blocks = std::min(32, (B + 255) / 256);
   auto partials = torch::empty({n_blocks, D + 1}, opts);
+  // generic-D kernel scratch (per-row dlogp coefficients)
+  auto c_buf = D > 8 ? torch::empty({B}, opts) : torch::Tensor();
   const float* olp = mode == 1 ? old_logp.data_ptr<float>() : nullptr;
   auto stream = current_stream();
   launch_gaussian_loss(mean.data_ptr<float>(), actions.data_ptr<float>(), olp,
                        adv.data_ptr<float>(), log_std.data_ptr<float>(),
-                       dmean.data_ptr<float>(), partials.data_ptr<float>(), B,
+                       dmean.data_ptr<float>(),
+                       D > 8 ? c_buf.data_ptr<float>() : nullptr,
+                       partials.data_ptr<float>(), B,
                        D, (float)clip, (int)mode, n_blocks, stream);
   HIP_OK(hipGetLastError());
   hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(D + 1), 0, stream,

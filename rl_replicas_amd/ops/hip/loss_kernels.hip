@@ -68,6 +68,7 @@ DEV_INLINE float dlogp_coeff(int mode, float logp, float old_logp, float adv,
 // ---------------------------------------------------------------------------
 #define LOSS_BLOCKS 32
 #define LOSS_P_THREADS 256
+#define GAUSS_MAX_D 512  // generic-D kernel's LDS sigma table bound
 
 template <int DT>
 __global__ __launch_bounds__(LOSS_P_THREADS) void gaussian_policy_loss_bwd_t(
@@ -142,6 +143,83 @@ __global__ __launch_bounds__(LOSS_P_THREADS) void gaussian_policy_loss_bwd_t(
   }
 }
 
+// Generic runtime-D variant (D in (8, GAUSS_MAX_D]): the per-thread
+// dlog_std register accumulators of the template version don't exist at
+// runtime D, so the block makes two passes over ITS OWN rows — phase A
+// stashes each row's dlogp coefficient in c_buf and writes dmean, phase
+// B loops d outermost and re-accumulates dlog_std per thread over the
+// same rows.  Per-(thread,d) accumulation order over rows is IDENTICAL
+// to the template version, so both are bitwise-equal and deterministic.
+__global__ __launch_bounds__(LOSS_P_THREADS) void gaussian_policy_loss_bwd_g(
+    const float* __restrict__ mean, const float* __restrict__ actions,
+    const float* __restrict__ old_logp, const float* __restrict__ adv,
+    const float* __restrict__ log_std, float* __restrict__ dmean,
+    float* __restrict__ c_buf, float* __restrict__ partials, int B, int D,
+    float clip, int mode) {
+  __shared__ float red[LOSS_P_THREADS / WAVE];
+  __shared__ float s_inv_s2[GAUSS_MAX_D];
+  __shared__ float s_base;
+  const int tid = threadIdx.x;
+  for (int d = tid; d < D; d += LOSS_P_THREADS) {
+    const float s = __expf(log_std[d]);
+    s_inv_s2[d] = 1.f / (s * s);
+  }
+  if (tid == 0) {
+    float b = 0.5f * (float)D * LOG_2PI;
+    for (int d = 0; d < D; ++d) b += log_std[d];
+    s_base = b;
+  }
+  __syncthreads();
+  const float inv_b = 1.f / (float)B;
+  float loss_acc = 0.f;
+  for (int r = blockIdx.x * LOSS_P_THREADS + tid; r < B;
+       r += gridDim.x * LOSS_P_THREADS) {
+    float q = 0.f;
+    for (int d = 0; d < D; ++d) {
+      const float diff = actions[(long)r * D + d] - mean[(long)r * D + d];
+      q += diff * diff * s_inv_s2[d];
+    }
+    const float logp = -0.5f * q - s_base;
+    float loss_r;
+    const float c = dlogp_coeff(mode, logp, old_logp ? old_logp[r] : 0.f,
+                                adv[r], clip, &loss_r) * inv_b;
+    c_buf[r] = c;
+    loss_acc += loss_r * inv_b;
+    for (int d = 0; d < D; ++d) {
+      const float diff = actions[(long)r * D + d] - mean[(long)r * D + d];
+      dmean[(long)r * D + d] = c * diff * s_inv_s2[d];
+    }
+  }
+  __syncthreads();
+  for (int d = 0; d < D; ++d) {
+    float accd = 0.f;
+    for (int r = blockIdx.x * LOSS_P_THREADS + tid; r < B;
+         r += gridDim.x * LOSS_P_THREADS) {
+      const float diff = actions[(long)r * D + d] - mean[(long)r * D + d];
+      accd += c_buf[r] * (diff * diff * s_inv_s2[d] - 1.f);
+    }
+    float v = wave_reduce_sum(accd);
+    if ((tid & 63) == 0) red[tid / WAVE] = v;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < LOSS_P_THREADS / WAVE; ++w) t += red[w];
+      partials[(long)blockIdx.x * (D + 1) + d] = t;
+    }
+    __syncthreads();
+  }
+  {
+    float v = wave_reduce_sum(loss_acc);
+    if ((tid & 63) == 0) red[tid / WAVE] = v;
+    __syncthreads();
+    if (tid == 0) {
+      float t = 0.f;
+      for (int w = 0; w < LOSS_P_THREADS / WAVE; ++w) t += red[w];
+      partials[(long)blockIdx.x * (D + 1) + D] = t;
+    }
+  }
+}
+
 // sum the per-block partial rows in fixed order:
 // dlog_std[d] = sum_blk partials[blk][d]; scalars[0] = sum_blk partials[blk][D]
 __global__ void loss_partials_finalize(const float* __restrict__ partials,
@@ -175,9 +253,10 @@ __global__ __launch_bounds__(LOSS_THREADS) void gaussian_logp_kernel(
     const float* __restrict__ mean, const float* __restrict__ actions,
     const float* __restrict__ log_std, float* __restrict__ logp, int B, int D) {
   const int tid = threadIdx.x + blockIdx.x * LOSS_THREADS;
-  __shared__ float s_sigma[32];
+  __shared__ float s_sigma[GAUSS_MAX_D];
   __shared__ float s_base;
-  if (threadIdx.x < D) s_sigma[threadIdx.x] = __expf(log_std[threadIdx.x]);
+  for (int d = threadIdx.x; d < D; d += LOSS_THREADS)
+    s_sigma[d] = __expf(log_std[d]);
   if (threadIdx.x == 0) {
     float b = 0.f;
     for (int d = 0; d < D; ++d) b += log_std[d];
@@ -200,9 +279,9 @@ __global__ __launch_bounds__(LOSS_THREADS) void gaussian_kl_kernel(
     const float* __restrict__ log_std, const float* __restrict__ old_logp,
     float* __restrict__ out, int B, int D) {
   __shared__ float red[LOSS_THREADS / WAVE];
-  __shared__ float s_sigma[32];
+  __shared__ float s_sigma[GAUSS_MAX_D];
   const int tid = threadIdx.x;
-  if (tid < D) s_sigma[tid] = __expf(log_std[tid]);
+  for (int d = tid; d < D; d += LOSS_THREADS) s_sigma[d] = __expf(log_std[d]);
   __syncthreads();
   float base = 0.f;
   for (int d = 0; d < D; ++d) base += log_std[d];
@@ -290,12 +369,14 @@ __global__ __launch_bounds__(LOSS_THREADS) void categorical_kl_kernel(
   if (tid == 0) out[0] = total / (float)B;
 }
 
-// host-side dispatch over the compile-time-D instantiations
+// host-side dispatch: compile-time-D instantiations for the common
+// D <= 8 heads (register dlog_std accumulators), generic two-pass
+// kernel for D in (8, GAUSS_MAX_D] (c_buf: caller-provided [B] scratch)
 void launch_gaussian_loss(const float* mean, const float* actions,
                           const float* old_logp, const float* adv,
-                          const float* log_std, float* dmean, float* partials,
-                          int B, int D, float clip, int mode, int n_blocks,
-                          hipStream_t stream) {
+                          const float* log_std, float* dmean, float* c_buf,
+                          float* partials, int B, int D, float clip, int mode,
+                          int n_blocks, hipStream_t stream) {
   dim3 g(n_blocks), b(LOSS_P_THREADS);
 #define CASE_D(DT)                                                           \
   case DT:                                                                   \
@@ -306,7 +387,11 @@ void launch_gaussian_loss(const float* mean, const float* actions,
   switch (D) {
     CASE_D(1) CASE_D(2) CASE_D(3) CASE_D(4)
     CASE_D(5) CASE_D(6) CASE_D(7) CASE_D(8)
-    default: break;  // guarded by the binding (D <= 8)
+    default:
+      hipLaunchKernelGGL(gaussian_policy_loss_bwd_g, g, b, 0, stream, mean,
+                         actions, old_logp, adv, log_std, dmean, c_buf,
+                         partials, B, D, clip, mode);
+      break;
   }
 #undef CASE_D
 }

@@ -34,9 +34,12 @@ def eager_gaussian_loss(mean, actions, old_logp, adv, log_std, clip, mode):
 class TestGaussianLoss:
     @pytest.mark.parametrize("mode", [0, 1])
     @pytest.mark.parametrize("scale", [1.0, 0.01])
-    def test_grads_match_autograd(self, ext, mode, scale):
+    # D=6: templated register kernel; D=17/64: generic two-pass kernel
+    # (gaussian_policy_loss_bwd_g) — the gate lift of round-1 VERDICT #6
+    @pytest.mark.parametrize("D", [6, 17, 64])
+    def test_grads_match_autograd(self, ext, mode, scale, D):
         torch.manual_seed(0)
-        B, D = 4000, 6
+        B = 4000
         mean = (torch.randn(B, D, device="cuda") * scale).requires_grad_(True)
         log_std = (-0.5 * torch.ones(D, device="cuda")).requires_grad_(True)
         actions = torch.randn(B, D, device="cuda")
@@ -77,9 +80,10 @@ class TestGaussianLoss:
         torch.testing.assert_close(dmean, mean.grad, rtol=1e-5, atol=1e-8)
         torch.testing.assert_close(dlog_std, log_std.grad, rtol=1e-5, atol=1e-7)
 
-    def test_logp_and_kl(self, ext):
+    @pytest.mark.parametrize("D", [6, 48])
+    def test_logp_and_kl(self, ext, D):
         torch.manual_seed(2)
-        B, D = 1000, 6
+        B = 1000
         mean = torch.randn(B, D, device="cuda")
         log_std = -0.3 * torch.ones(D, device="cuda")
         actions = torch.randn(B, D, device="cuda")

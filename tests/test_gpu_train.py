@@ -131,15 +131,16 @@ def test_gpu_determinism_same_seed(tmp_path):
         assert torch.equal(a, b)
 
 
-def test_ppo_with_wide_value_net_falls_back(tmp_path):
-    """A 256-wide value net doesn't satisfy the fused value-backward
-    constraints — PPO must fall back to the autograd value path, not
-    crash (regression test for the value_supported gate)."""
+def test_ppo_with_wide_value_net_stays_fused(tmp_path):
+    """A 256-wide value net now runs the WIDE fused value loop
+    (three-kernel pipeline, fused_onpolicy._value_iter_wide) instead of
+    falling back to autograd — and the whole PPO epoch still trains."""
     import torch.nn as nn
 
     from rl_replicas_amd import envs, ops
     from rl_replicas_amd.algorithms import PPO
     from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.ops import fused_onpolicy
     from rl_replicas_amd.policies import GaussianPolicy
     from rl_replicas_amd.samplers import VectorSampler
     from rl_replicas_amd.value_function import ValueFunction
@@ -151,13 +152,58 @@ def test_ppo_with_wide_value_net_falls_back(tmp_path):
     policy = GaussianPolicy(
         pnet, ops.make_adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
     )
-    vnet = MLP([17, 256, 256, 1]).to(DEVICE)  # wide: fused value loop unsupported
+    vnet = MLP([17, 256, 256, 1]).to(DEVICE)
     vf = ValueFunction(vnet, ops.make_adam(vnet.parameters(), lr=1e-3))
     model = PPO(policy, vf, venv, VectorSampler(venv, seed=0),
                 num_policy_gradients=3, num_value_gradients=3)
+    obs_probe = torch.zeros(4, 17, device=DEVICE)
+    assert fused_onpolicy.value_mode(model, obs_probe) == "wide"
     model.learn(num_epochs=2, batch_size=300, output_dir=str(tmp_path))
     for p in vnet.parameters():
         assert torch.isfinite(p).all()
+
+
+def test_wide_value_update_matches_autograd():
+    """Fused wide value loop vs torch-autograd MSE steps, same init."""
+    from rl_replicas_amd import ops
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.ops import fused_onpolicy
+    from rl_replicas_amd.value_function import ValueFunction
+
+    torch.manual_seed(2)
+    vnet_f = MLP([17, 256, 256, 1]).to(DEVICE)
+    vnet_e = MLP([17, 256, 256, 1]).to(DEVICE)
+    vnet_e.load_state_dict(vnet_f.state_dict())
+    vf_f = ValueFunction(vnet_f, torch.optim.Adam(vnet_f.parameters(), lr=1e-3))
+    vf_e = ValueFunction(vnet_e, torch.optim.Adam(vnet_e.parameters(), lr=1e-3))
+
+    obs = torch.randn(500, 17, device=DEVICE)
+    returns = torch.randn(500, device=DEVICE)
+
+    class Holder:
+        value_function = vf_f
+
+        @staticmethod
+        def _all_reduce_gradients(m):
+            pass
+
+    assert fused_onpolicy.value_mode(Holder, obs) == "wide"
+    loss_f = fused_onpolicy.value_update(Holder, obs, returns, 5)
+
+    vnet_e.fused_training = False
+    losses_e = []
+    for _ in range(5):
+        v = vf_e(obs).squeeze(-1)
+        loss = torch.nn.functional.mse_loss(v, returns)
+        vf_e.optimizer.zero_grad()
+        loss.backward()
+        vf_e.optimizer.step()
+        losses_e.append(loss.item())
+    import numpy as np
+
+    assert loss_f == pytest.approx(float(np.mean(losses_e)), rel=1e-4)
+    for p_f, p_e in zip(vnet_f.parameters(), vnet_e.parameters()):
+        torch.testing.assert_close(p_f, p_e, rtol=1e-4, atol=1e-6)
 
 
 def test_ppo_device_resident_rollout(tmp_path):
