@@ -30,15 +30,16 @@ def _worker_allreduce(rank: int, world: int, tmpdir: str):
     torch.manual_seed(0)  # identical model on all ranks
     from rl_replicas_amd.parallel import all_reduce_gradients, global_normalize
 
+    rows = 2  # rows per rank
     net = torch.nn.Linear(4, 3)
-    full_x = torch.randn(8, 4, generator=torch.Generator().manual_seed(42))
-    full_y = torch.randn(8, 3, generator=torch.Generator().manual_seed(43))
+    full_x = torch.randn(rows * world, 4, generator=torch.Generator().manual_seed(42))
+    full_y = torch.randn(rows * world, 3, generator=torch.Generator().manual_seed(43))
     # single-process oracle gradient on the full batch
     loss_full = torch.nn.functional.mse_loss(net(full_x), full_y)
     oracle = torch.autograd.grad(loss_full, list(net.parameters()))
 
-    # each rank computes grads on its half, then all-reduces
-    shard = slice(rank * 4, (rank + 1) * 4)
+    # each rank computes grads on its shard, then all-reduces
+    shard = slice(rank * rows, (rank + 1) * rows)
     loss = torch.nn.functional.mse_loss(net(full_x[shard]), full_y[shard])
     net.zero_grad()
     loss.backward()
@@ -47,10 +48,11 @@ def _worker_allreduce(rank: int, world: int, tmpdir: str):
         torch.testing.assert_close(p.grad, g_oracle, rtol=1e-5, atol=1e-6)
 
     # global normalization == normalizing the concatenated vector
-    full_v = torch.randn(10, generator=torch.Generator().manual_seed(7))
-    local = full_v[rank * 5 : (rank + 1) * 5]
+    n = 5
+    full_v = torch.randn(n * world, generator=torch.Generator().manual_seed(7))
+    local = full_v[rank * n : (rank + 1) * n]
     got = global_normalize(local)
-    expected = ((full_v - full_v.mean()) / full_v.std())[rank * 5 : (rank + 1) * 5]
+    expected = ((full_v - full_v.mean()) / full_v.std())[rank * n : (rank + 1) * n]
     torch.testing.assert_close(got, expected, rtol=1e-5, atol=1e-6)
     dist.destroy_process_group()
 
@@ -266,3 +268,23 @@ def _worker_td3_sync(rank: int, world: int, tmpdir: str):
 def test_two_rank_gloo(worker, tmp_path):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     mp.spawn(worker, args=(2, str(tmp_path)), nprocs=2, join=True)
+
+
+# rank-count invariance at the SCALE shapes (4 and 8 ranks = half/full
+# MI355X node); every algorithm's replicas must stay bitwise identical
+# and the reduced math must equal the full-batch oracle at any world
+# size (round-1 VERDICT "prove the RCCL path" item)
+@pytest.mark.parametrize(
+    "worker",
+    [_worker_allreduce, _worker_ppo_sync, _worker_ppo_device_sampler,
+     _worker_trpo_sync, _worker_ddpg_sync, _worker_td3_sync],
+)
+def test_four_rank_gloo(worker, tmp_path):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    mp.spawn(worker, args=(4, str(tmp_path)), nprocs=4, join=True)
+
+
+@pytest.mark.parametrize("worker", [_worker_allreduce, _worker_ppo_sync])
+def test_eight_rank_gloo(worker, tmp_path):
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    mp.spawn(worker, args=(8, str(tmp_path)), nprocs=8, join=True)
