@@ -19,11 +19,17 @@ class FusedAdam(torch.optim.Adam):
     """torch.optim.Adam with a fused multi-tensor HIP step."""
 
     @torch.no_grad()
-    def step(self, closure=None, *, step_delta: float = 0.0, do_bump: bool = True):
+    def step(self, closure=None, *, step_delta: float = 0.0, do_bump: bool = True,
+             gate=None):
         """step_delta/do_bump: captured-loop mode — iteration i of a
         hipGraph-captured loop passes step_delta=i, do_bump=False and the
         loop bumps once by num_iters at the end (bump_steps); bitwise
-        identical to per-iteration bumps for integer-valued fp32 steps."""
+        identical to per-iteration bumps for integer-valued fp32 steps.
+
+        gate: optional 1-elem fp32 device tensor; the update (and bump)
+        are skipped while it is 0 — lets a captured loop realize data-
+        dependent early stopping (PPO's KL test) without host control
+        flow."""
         loss = None
         if closure is not None:
             with torch.enable_grad():
@@ -70,6 +76,7 @@ class FusedAdam(torch.optim.Adam):
                 float(group.get("weight_decay", 0.0)),
                 float(step_delta),
                 bool(do_bump),
+                gate,
             )
         return loss
 

@@ -219,13 +219,22 @@ class _CapturedLoop:
 
 
 class _GraphedPPO:
-    """One captured PPO policy iteration: fwd + loss + bwd + Adam +
-    KL-eval forward, replayed up to num_policy_gradients times.
+    """The PPO policy-update loop on hipGraphs.
 
-    `split=True` (data parallelism) captures TWO graphs — grads before
-    / Adam+KL after — and runs the eager gradient all-reduce between
-    them on the capture-stable .grad buffers; collectives are never
-    captured."""
+    Single-process (`split=False`): the WHOLE num_policy_gradients loop
+    is ONE captured graph per epoch.  The reference's KL early stop
+    (ppo.py:173-181) is host control flow; here it becomes a device-side
+    gate — each iteration's Adam step is skipped once the post-step
+    approximate KL has exceeded 1.5*max_kl (fused_adam_kernel's gate
+    pointer), so later iterations replay as frozen no-ops and the
+    resulting parameters equal the break-based loop's.  One replay +
+    two scalar readbacks per epoch instead of 80 replays with a KL sync
+    each.
+
+    `split=True` (data parallelism) captures TWO graphs per iteration —
+    grads before / Adam+KL after — and runs the eager gradient
+    all-reduce between them on the capture-stable .grad buffers;
+    collectives (and the rank-synchronized KL decision) stay eager."""
 
     def __init__(self, algo, kind: str, obs0: Tensor, actions0: Tensor,
                  adv0: Tensor, old_logp0: Tensor, split: bool):
@@ -269,28 +278,51 @@ class _GraphedPPO:
                 param.grad = grad
             return scalars
 
-        def body_post():
-            policy.optimizer.step()
+        def kl_eval():
             new_out = _forward_only(mlp, self.obs)
             if kind == "gaussian":
-                kl = ext.gaussian_kl(new_out, self.actions, policy.log_std.data,
-                                     self.old_logp)
-            else:
-                kl = ext.categorical_kl(new_out, self.actions, self.old_logp)
-            return kl
+                return ext.gaussian_kl(new_out, self.actions,
+                                       policy.log_std.data, self.old_logp)
+            return ext.categorical_kl(new_out, self.actions, self.old_logp)
 
         state = [p.data for p in policy.parameters()]
         state += _ensure_adam_state(policy.optimizer)
         if split:
+            def body_post():
+                policy.optimizer.step()
+                return kl_eval()
+
             # order matters: pre's capture pins the .grad buffers that
             # post's Adam capture reads
             self.pre = _CapturedLoop(body_pre, state)
             self.post = _CapturedLoop(body_post, state)
         else:
+            dev = obs0.device
+            num_iters = int(algo.num_policy_gradients)
+            thr = 1.5 * float(algo.max_kl_divergence)
+            self.gate = torch.ones(1, device=dev)
+            self.kl_final = torch.zeros(1, device=dev)
+            self.loss0 = torch.zeros(1, device=dev)
+            self.iters_done = torch.zeros(1, device=dev)
+
             def body():
-                scalars = body_pre()
-                kl = body_post()
-                return scalars, kl
+                self.gate.fill_(1.0)
+                self.iters_done.zero_()
+                for i in range(num_iters):
+                    scalars = body_pre()
+                    if i == 0:
+                        self.loss0.copy_(scalars[:1])
+                    executed = self.gate.clone()  # 1 while still active
+                    policy.optimizer.step(gate=self.gate)
+                    kl = kl_eval()[:1]
+                    # the reported KL is the last one computed while
+                    # active (the stop-triggering value, or the final
+                    # iteration's) — reference ppo.py:176-181
+                    self.kl_final.copy_(
+                        torch.where(executed.bool(), kl, self.kl_final)
+                    )
+                    self.iters_done.add_(executed)
+                    self.gate.mul_((kl <= thr).float())
 
             self.loop = _CapturedLoop(body, state)
 
@@ -300,15 +332,24 @@ class _GraphedPPO:
         self.adv.copy_(advantages)
         self.old_logp.copy_(old_logp)
         policy = algo.policy
+        if not self.split:
+            self.loop.replay()
+            iters = int(self.iters_done)
+            if iters < algo.num_policy_gradients:
+                logger.info(
+                    "Early stopping at update %d due to reaching max KL divergence.",
+                    iters - 1,
+                )
+            return {
+                "policy/loss": float(self.loss0[0]),
+                "policy/kl_divergence": float(self.kl_final[0]),
+            }
         loss_before = None
         approximate_kl = 0.0
         for i in range(algo.num_policy_gradients):
-            if self.split:
-                scalars = self.pre.replay()
-                algo._all_reduce_gradients(policy)
-                kl = self.post.replay()
-            else:
-                scalars, kl = self.loop.replay()
+            scalars = self.pre.replay()
+            algo._all_reduce_gradients(policy)
+            kl = self.post.replay()
             if loss_before is None:
                 loss_before = float(scalars[0])
             approximate_kl = float(algo._reduce_scalar_mean(kl[0]))

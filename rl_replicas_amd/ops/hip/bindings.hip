@@ -686,7 +686,15 @@ void fused_adam_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
                  std::vector<torch::Tensor> exp_avg_sqs,
                  std::vector<torch::Tensor> steps, double lr, double beta1,
                  double beta2, double eps, double weight_decay,
-                 double step_delta, bool do_bump) {
+                 double step_delta, bool do_bump,
+                 c10::optional<torch::Tensor> gate) {
+  const float* gate_ptr = nullptr;
+  if (gate.has_value()) {
+    TORCH_CHECK(gate->scalar_type() == torch::kFloat32 && gate->is_cuda() &&
+                    gate->numel() == 1,
+                "gate must be a 1-element fp32 CUDA tensor");
+    gate_ptr = gate->data_ptr<float>();
+  }
   size_t n = params.size();
   for (size_t base = 0; base < n; base += MT_MAX_TENSORS) {
     AdamArgs a{};
@@ -706,6 +714,7 @@ void fused_adam_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
     a.eps = (float)eps;
     a.weight_decay = (float)weight_decay;
     a.step_delta = (float)step_delta;
+    a.gate = gate_ptr;
     int max_chunks = 1;
     for (int i = 0; i < a.n_tensors; ++i)
       max_chunks = std::max(max_chunks, (a.numel[i] + 8191) / 8192);
@@ -985,7 +994,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("exp_avg_sqs"), py::arg("steps"), py::arg("lr"),
         py::arg("beta1"), py::arg("beta2"), py::arg("eps"),
         py::arg("weight_decay"), py::arg("step_delta") = 0.0,
-        py::arg("do_bump") = true);
+        py::arg("do_bump") = true, py::arg("gate") = py::none());
   m.def("fused_polyak_", &fused_polyak_, "fused multi-tensor Polyak (gfx950)");
   m.def("gaussian_policy_loss", &gaussian_policy_loss,
         "fused Gaussian VPG/PPO loss fwd+bwd (gfx950)");

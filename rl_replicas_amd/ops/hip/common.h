@@ -104,6 +104,9 @@ struct AdamArgs {
   // bakes per-iteration deltas so ONE bump per graph suffices); also the
   // bump amount for adam_step_bump_kernel
   float step_delta;
+  // optional device gate: update and bump are skipped while *gate == 0
+  // (captured PPO policy loop: KL early stop without host control flow)
+  const float* gate;
 };
 
 struct PolyakArgs {

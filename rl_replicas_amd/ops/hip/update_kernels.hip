@@ -17,6 +17,9 @@
 #define MT_CHUNK 8192  // elements per (tensor, chunk) block
 
 __global__ __launch_bounds__(256) void fused_adam_kernel(AdamArgs a) {
+  // device-side gate: a captured loop can disable later updates without
+  // host control flow (PPO's KL early stop runs entirely on device)
+  if (a.gate && *a.gate == 0.f) return;
   const int t = blockIdx.x;
   if (t >= a.n_tensors) return;
   const int n = a.numel[t];
@@ -47,6 +50,7 @@ __global__ __launch_bounds__(256) void fused_adam_kernel(AdamArgs a) {
 // bump every tensor's step counter once per optimizer step (launched
 // after fused_adam_kernel on the same stream)
 __global__ void adam_step_bump_kernel(AdamArgs a) {
+  if (a.gate && *a.gate == 0.f) return;
   const int t = threadIdx.x;
   if (t < a.n_tensors) a.step[t][0] += a.step_delta;  // 1.0 for a plain step
 }

@@ -271,6 +271,40 @@ class TestGraphedLoops:
         for p1, p2 in zip(m1.value_function.parameters(), m2.value_function.parameters()):
             torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-7)
 
+    def test_graphed_early_stop_gate_matches_eager(self, ext, monkeypatch):
+        """KL early stop inside the SINGLE captured graph (device gate on
+        the Adam kernel) must stop at the same iteration as the eager
+        break-based loop: same final params, same reported KL."""
+        from rl_replicas_amd.ops import fused_onpolicy as fop
+        from rl_replicas_amd.ops.fused_adam import FusedAdam
+
+        torch.manual_seed(43)
+        obs = torch.randn(1000, 17, device="cuda")
+        actions = torch.randn(1000, 6, device="cuda")
+        adv = 50.0 * torch.randn(1000, device="cuda")  # big steps -> KL blows past thr
+
+        m1 = self._make_ppo(3)
+        m2 = self._make_ppo(3)
+        m1.max_kl_divergence = 1e-4  # force an early stop within 5 iters
+        m2.max_kl_divergence = 1e-4
+        assert isinstance(m1.policy.optimizer, FusedAdam)
+
+        monkeypatch.setenv("RL_REPLICAS_AMD_DISABLE_GRAPHS", "1")
+        r_eager = fop.ppo_update(m2, obs, actions, adv)
+        monkeypatch.delenv("RL_REPLICAS_AMD_DISABLE_GRAPHS")
+        r_graph = fop.ppo_update(m1, obs, actions, adv)
+
+        # the stop must actually have triggered for this to test anything
+        graphed = m1._ppo_policy_graph[1]
+        assert int(graphed.iters_done) < m1.num_policy_gradients
+        assert abs(r_graph["policy/kl_divergence"] - r_eager["policy/kl_divergence"]) < 1e-5
+        for p1, p2 in zip(m1.policy.parameters(), m2.policy.parameters()):
+            torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-7)
+        # Adam step counters must have advanced only for executed iters
+        s1 = m1.policy.optimizer.state[next(iter(m1.policy.parameters()))]["step"]
+        s2 = m2.policy.optimizer.state[next(iter(m2.policy.parameters()))]["step"]
+        assert float(s1) == float(s2)
+
     def test_graph_replay_across_epochs(self, ext):
         """Second epoch reuses the cached graph with new data; params
         keep evolving and stay finite."""
