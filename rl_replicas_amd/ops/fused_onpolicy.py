@@ -305,26 +305,37 @@ class _GraphedPPO:
             self.loss0 = torch.zeros(1, device=dev)
             self.iters_done = torch.zeros(1, device=dev)
 
-            def body():
-                self.gate.fill_(1.0)
-                self.iters_done.zero_()
-                for i in range(num_iters):
-                    scalars = body_pre()
-                    if i == 0:
-                        self.loss0.copy_(scalars[:1])
-                    executed = self.gate.clone()  # 1 while still active
-                    policy.optimizer.step(gate=self.gate)
-                    kl = kl_eval()[:1]
-                    # the reported KL is the last one computed while
-                    # active (the stop-triggering value, or the final
-                    # iteration's) — reference ppo.py:176-181
-                    self.kl_final.copy_(
-                        torch.where(executed.bool(), kl, self.kl_final)
-                    )
-                    self.iters_done.add_(executed)
-                    self.gate.mul_((kl <= thr).float())
+            def make_chunk(start: int, count: int):
+                def chunk():
+                    if start == 0:
+                        self.gate.fill_(1.0)
+                        self.iters_done.zero_()
+                    for i in range(start, start + count):
+                        scalars = body_pre()
+                        if i == 0:
+                            self.loss0.copy_(scalars[:1])
+                        policy.optimizer.step(gate=self.gate)
+                        kl = kl_eval()
+                        # one-kernel bookkeeping: while active, record the
+                        # KL (the reported value is the stop-triggering
+                        # one, reference ppo.py:176-181) and the iteration
+                        # count, then close the gate if KL > thr
+                        ext.ppo_gate_update_(self.gate, kl, self.kl_final,
+                                             self.iters_done, thr)
 
-            self.loop = _CapturedLoop(body, state)
+                return chunk
+
+            # chunked capture: the KL early stop is common in steady
+            # state, and a gate-frozen iteration still executes its
+            # kernels — chunks bound that waste to CHUNK-1 iterations
+            # while keeping host syncs at one gate readback per chunk
+            # (vs per-iteration replay+sync, or one full-loop graph
+            # that always runs all 80)
+            CHUNK = 10
+            self.chunks = []
+            for start in range(0, num_iters, CHUNK):
+                count = min(CHUNK, num_iters - start)
+                self.chunks.append(_CapturedLoop(make_chunk(start, count), state))
 
     def run(self, algo, obs, actions, advantages, old_logp) -> Dict[str, float]:
         self.obs.copy_(obs)
@@ -333,7 +344,10 @@ class _GraphedPPO:
         self.old_logp.copy_(old_logp)
         policy = algo.policy
         if not self.split:
-            self.loop.replay()
+            for chunk in self.chunks:
+                chunk.replay()
+                if float(self.gate[0]) == 0.0:
+                    break
             iters = int(self.iters_done)
             if iters < algo.num_policy_gradients:
                 logger.info(

@@ -103,6 +103,9 @@ __global__ void replay_gather_kernel(ReplayGatherArgs a);
 __global__ void fused_adam_kernel(AdamArgs a);
 __global__ void adam_step_bump_kernel(AdamArgs a);
 __global__ void fused_polyak_kernel(PolyakArgs a);
+__global__ void ppo_gate_update_kernel(float* gate, const float* kl,
+                                       float* kl_final, float* iters_done,
+                                       float thr);
 
 namespace {
 
@@ -947,6 +950,16 @@ std::vector<torch::Tensor> replay_gather(torch::Tensor obs, torch::Tensor act,
   return {qin, obs_out, nxt_out, rew_out, dn_out};
 }
 
+void ppo_gate_update_(torch::Tensor gate, torch::Tensor kl,
+                      torch::Tensor kl_final, torch::Tensor iters_done,
+                      double thr) {
+  hipLaunchKernelGGL(ppo_gate_update_kernel, dim3(1), dim3(1), 0,
+                     current_stream(), gate.data_ptr<float>(),
+                     kl.data_ptr<float>(), kl_final.data_ptr<float>(),
+                     iters_done.data_ptr<float>(), (float)thr);
+  HIP_OK(hipGetLastError());
+}
+
 void counter_add_(torch::Tensor ctr, int64_t delta) {
   TORCH_CHECK(ctr.scalar_type() == torch::kInt64 && ctr.is_cuda() && ctr.numel() == 1,
               "ctr must be a 1-element int64 CUDA tensor");
@@ -975,6 +988,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("value_loss_finalize", &value_loss_finalize,
         "batched deferred value-loss finalize (gfx950)");
   m.def("adam_bump_", &adam_bump_, "bump Adam step counters (gfx950)");
+  m.def("ppo_gate_update_", &ppo_gate_update_,
+        "device-side KL early-stop gate bookkeeping (gfx950)");
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");
   m.def("normalize", &normalize, "fused mean/std normalize (gfx950)");
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");

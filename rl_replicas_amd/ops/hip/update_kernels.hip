@@ -55,6 +55,20 @@ __global__ void adam_step_bump_kernel(AdamArgs a) {
   if (t < a.n_tensors) a.step[t][0] += a.step_delta;  // 1.0 for a plain step
 }
 
+// One-thread bookkeeping for the captured PPO policy loop's device-side
+// KL early stop: while the gate is open, record the KL and iteration
+// count; close the gate when the KL crosses the threshold.  Replaces a
+// ~6-torch-op chain per captured iteration.
+__global__ void ppo_gate_update_kernel(float* gate, const float* kl,
+                                       float* kl_final, float* iters_done,
+                                       float thr) {
+  if (*gate != 0.f) {
+    *kl_final = *kl;
+    *iters_done += 1.f;
+    if (*kl > thr) *gate = 0.f;
+  }
+}
+
 __global__ __launch_bounds__(256) void fused_polyak_kernel(PolyakArgs a) {
   const int t = blockIdx.x;
   if (t >= a.n_tensors) return;
