@@ -57,11 +57,29 @@ class TestGaussianLoss:
             log_std.detach().contiguous(), 0.2, mode,
         )
         torch.testing.assert_close(scalars[0], loss.detach(), rtol=1e-4, atol=1e-6)
-        # wider heads sum more fp32 terms per row than torch's order:
-        # observed max drift ~1e-3 relative at D=64 (no boundary flips)
+        # wider heads sum more fp32 terms per row than torch's order
         rtol, atol = (2e-3, 1e-5) if D > 8 else (1e-4, 1e-7)
-        torch.testing.assert_close(dmean, mean.grad, rtol=rtol, atol=atol)
-        torch.testing.assert_close(dlog_std, log_std.grad, rtol=rtol, atol=max(atol, 1e-6))
+        # rows whose ratio sits ON a clip boundary are excluded from the
+        # strict comparison: a 1-ulp logp disagreement flips the clip
+        # indicator there, making torch's own gradient discontinuous (at
+        # D=64 one of 4000 rows hits this; both answers are "correct")
+        if mode == 1:
+            with torch.no_grad():
+                logp_d = Independent(
+                    Normal(mean.detach(), torch.exp(log_std.detach())), 1
+                ).log_prob(actions)
+                ratio = torch.exp(logp_d - old_logp)
+                boundary = ((ratio - 0.8).abs() < 1e-4) | ((ratio - 1.2).abs() < 1e-4)
+        else:
+            boundary = torch.zeros(B, dtype=torch.bool, device="cuda")
+        ok = ~boundary
+        n_flipped = int(boundary.sum())
+        assert n_flipped < 5  # boundary rows must stay rare
+        torch.testing.assert_close(dmean[ok], mean.grad[ok], rtol=rtol, atol=atol)
+        # dlog_std aggregates over ALL rows, boundary ones included:
+        # each flipped row can shift it by ~|adv|/B * (z^2-1) ~ 5e-3
+        atol_ls = max(atol, 1e-6) + 5e-3 * n_flipped
+        torch.testing.assert_close(dlog_std, log_std.grad, rtol=rtol, atol=atol_ls)
 
     def test_exact_tie_first_iteration(self, ext):
         """First PPO iteration: ratio == 1 everywhere (old == new).
@@ -281,12 +299,16 @@ class TestGraphedLoops:
         torch.manual_seed(43)
         obs = torch.randn(1000, 17, device="cuda")
         actions = torch.randn(1000, 6, device="cuda")
-        adv = 50.0 * torch.randn(1000, device="cuda")  # big steps -> KL blows past thr
+        adv = torch.randn(1000, device="cuda")
 
         m1 = self._make_ppo(3)
         m2 = self._make_ppo(3)
-        m1.max_kl_divergence = 1e-4  # force an early stop within 5 iters
-        m2.max_kl_divergence = 1e-4
+        for m in (m1, m2):
+            # guaranteed stop after iteration 0 (any finite KL > 1.5*-1):
+            # exercises gate-masked iterations 1-9 of chunk 0 AND the
+            # chunk-break (chunks 1-2 never replay)
+            m.max_kl_divergence = -1.0
+            m.num_policy_gradients = 25
         assert isinstance(m1.policy.optimizer, FusedAdam)
 
         monkeypatch.setenv("RL_REPLICAS_AMD_DISABLE_GRAPHS", "1")
@@ -294,9 +316,8 @@ class TestGraphedLoops:
         monkeypatch.delenv("RL_REPLICAS_AMD_DISABLE_GRAPHS")
         r_graph = fop.ppo_update(m1, obs, actions, adv)
 
-        # the stop must actually have triggered for this to test anything
         graphed = m1._ppo_policy_graph[1]
-        assert int(graphed.iters_done) < m1.num_policy_gradients
+        assert int(graphed.iters_done) == 1  # stopped after the first update
         assert abs(r_graph["policy/kl_divergence"] - r_eager["policy/kl_divergence"]) < 1e-5
         for p1, p2 in zip(m1.policy.parameters(), m2.policy.parameters()):
             torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-7)
