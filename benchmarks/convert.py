@@ -41,6 +41,24 @@ def read_metric_curve(metrics_csv: str, tag: str) -> Dict[int, float]:
     return curve
 
 
+def read_tfevents_curve(tb_dir: str, tag: str) -> Dict[int, float]:
+    """Read a tag's scalar curve from events.out.tfevents.* files
+    (reference convert.py reads tfevents via EventAccumulator; this
+    uses the native reader in rl_replicas_amd.tfevents)."""
+    from rl_replicas_amd.tfevents import read_scalar_events
+
+    curve: Dict[int, float] = {}
+    if not os.path.isdir(tb_dir):
+        return curve
+    for fname in sorted(os.listdir(tb_dir)):
+        if not fname.startswith("events.out.tfevents"):
+            continue
+        for ev_tag, value, step in read_scalar_events(os.path.join(tb_dir, fname)):
+            if ev_tag == tag:
+                curve[step] = value
+    return curve
+
+
 def rolling_mean(values: np.ndarray, window: int) -> np.ndarray:
     out = np.empty_like(values, dtype=np.float64)
     for i in range(len(values)):
@@ -61,6 +79,14 @@ def convert_env(env_dir: str, env_id: str, outdir: str) -> None:
             path = os.path.join(algo_dir, seed_dir, "metrics.csv")
             if os.path.exists(path):
                 seed_curves.append(read_metric_curve(path, tag))
+            else:
+                # tfevents input path (the reference converter's native
+                # format; written here by rl_replicas_amd/tfevents.py)
+                curve = read_tfevents_curve(
+                    os.path.join(algo_dir, seed_dir, "tensorboard"), tag
+                )
+                if curve:
+                    seed_curves.append(curve)
         if not seed_curves:
             continue
         common_steps = sorted(set.intersection(*(set(c) for c in seed_curves)))

@@ -10,11 +10,14 @@ including the reference's historical `avarage` spellings) are emitted
 by the algorithms unchanged so downstream tooling keyed on the
 reference's tags keeps working.
 
-This stack has no tensorboard package, so scalars marked
-`tensorboard=True` are additionally persisted to a CSV file
-(`<log_dir>/metrics.csv`: tag,step,value) — the benchmark converter
-(benchmarks/convert.py analogue) reads that.  If tensorboard IS
-importable, a SummaryWriter is used as well.
+Scalars marked `tensorboard=True` are persisted to BOTH a CSV file
+(`<log_dir>/metrics.csv`: tag,step,value — read by the benchmark
+converter) and a real `events.out.tfevents.*` file under
+`<log_dir>/tensorboard` written by the native tfevents writer
+(rl_replicas_amd/tfevents.py) — so downstream TensorBoard tooling
+keyed on the reference's event files keeps working even though the
+tensorboard package isn't installed here.  If tensorboard IS
+importable, its SummaryWriter is used instead of the native writer.
 
 MI355X addition: `record_phase_ms` aggregates per-phase HIP-event /
 wall timings (sample / h2d / forward / backward / allreduce / optimizer)
@@ -48,9 +51,13 @@ class MetricsManager:
         self._csv = csv.writer(self._csv_file)
         if self._csv_file.tell() == 0:
             self._csv.writerow(["tag", "step", "value"])
-        self.tensorboard_writer = None
+        tb_dir = os.path.join(log_dir, "tensorboard")
         if _HAS_TB:
-            self.tensorboard_writer = SummaryWriter(os.path.join(log_dir, "tensorboard"))
+            self.tensorboard_writer = SummaryWriter(tb_dir)
+        else:
+            from rl_replicas_amd.tfevents import EventFileWriter
+
+            self.tensorboard_writer = EventFileWriter(tb_dir)
         self._phase_ms: Dict[str, float] = {}
 
     def record_scalar(
