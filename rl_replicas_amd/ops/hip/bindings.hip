@@ -217,7 +217,11 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
   pick_tile(args.batch, max_width, &rows, &maxw);
 
   auto opts = x.options();
-  const bool wide = maxw == 256;
+  // per-layer column-split kernels need enough rows to fill the chip
+  // row-wise too; at minibatch scale (DDPG/TD3: 100 rows) the whole-net
+  // <32,256> fused kernel replaces L launches with 1 (weights read
+  // directly from global, wstage mode 2)
+  const bool wide = maxw == 256 && args.batch > 512;
   std::vector<torch::Tensor> outs;  // [final, h0..h_{L-2}]
   torch::Tensor final_out = torch::empty({x.size(0), args.dims[L]}, opts);
   outs.push_back(final_out);
