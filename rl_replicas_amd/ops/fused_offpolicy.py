@@ -261,39 +261,83 @@ class _GraphedOffPolicy:
         polyak_src = [p.data for p in polyak_src]
         polyak_dst = [p.data for p in polyak_dst]
 
+        import os
+
+        use_streams = os.environ.get("RL_REPLICAS_AMD_OFFPOLICY_STREAMS", "1") != "0"
+        # side streams for the independent work inside one iteration:
+        # the two critics' chains and the target chain have no mutual
+        # data dependence until the targets/losses join.  Captured
+        # wait_stream edges become graph dependencies, so the replay
+        # overlaps these small (16-WG) kernels instead of serializing
+        # every launch on one stream.
+        s1 = torch.cuda.Stream() if use_streams else None
+        s2 = torch.cuda.Stream() if use_streams else None
+
         def body():
             pi_k = 0
+            main = torch.cuda.current_stream()
             for i in range(num_iters):
                 qin, obs, nxt, rew, dn = ext.replay_gather(
                     st["observations"], act_flat, st["rewards"],
                     st["next_observations"], st["dones"], buf._size_dev,
                     minibatch_size, seed, 3 * i, self.ctr,
                 )
-                # pre-update Q values for the epoch metrics
-                # (reference td3.py:230-236)
-                for q, buf_q in zip(q_fns, self.all_q):
-                    buf_q[i].copy_(forward_only(q, qin).view(-1))
-                # target chain
-                na = forward_only(algo.target_policy, nxt)
-                if twin:
-                    na = ext.td3_smooth(
-                        na, seed, 3 * i + 1,
-                        float(algo.target_noise_scale),
-                        float(algo.target_noise_clip), action_limit, self.ctr,
-                    )
-                qt_in = torch.cat([nxt, na], dim=-1).contiguous()
-                if twin:
-                    q1t = forward_only(tq_fns[0], qt_in).view(-1)
-                    q2t = forward_only(tq_fns[1], qt_in).view(-1)
-                    targets = ext.q_target_min2(rew, dn, q1t, q2t, gamma)
-                else:
+
+                def target_chain():
+                    na = forward_only(algo.target_policy, nxt)
+                    if twin:
+                        na2 = ext.td3_smooth(
+                            na, seed, 3 * i + 1,
+                            float(algo.target_noise_scale),
+                            float(algo.target_noise_clip), action_limit,
+                            self.ctr,
+                        )
+                    else:
+                        na2 = na
+                    qt_in = torch.cat([nxt, na2], dim=-1).contiguous()
+                    if twin:
+                        q1t = forward_only(tq_fns[0], qt_in).view(-1)
+                        q2t = forward_only(tq_fns[1], qt_in).view(-1)
+                        return ext.q_target_min2(rew, dn, q1t, q2t, gamma)
                     qt = forward_only(tq_fns[0], qt_in).view(-1)
-                    targets = ext.q_target(rew, dn, qt, gamma)
-                for q, buf_l in zip(q_fns, self.q_losses):
-                    buf_l[i].copy_(
-                        q_step(q, obs, None, targets, _noop_hook, qin=qin,
-                               step_delta=float(i))
+                    return ext.q_target(rew, dn, qt, gamma)
+
+                if use_streams:
+                    s1.wait_stream(main)
+                    s2.wait_stream(main)
+                    # main: q1 logging fwd (reference td3.py:230-236)
+                    self.all_q[0][i].copy_(forward_only(q_fns[0], qin).view(-1))
+                    with torch.cuda.stream(s1):
+                        if twin:
+                            self.all_q[1][i].copy_(
+                                forward_only(q_fns[1], qin).view(-1)
+                            )
+                    with torch.cuda.stream(s2):
+                        targets = target_chain()
+                    main.wait_stream(s1)
+                    main.wait_stream(s2)
+                    # critic steps in parallel: q1 on main, q2 on s1
+                    s1.wait_stream(main)
+                    self.q_losses[0][i].copy_(
+                        q_step(q_fns[0], obs, None, targets, _noop_hook,
+                               qin=qin, step_delta=float(i))
                     )
+                    if twin:
+                        with torch.cuda.stream(s1):
+                            self.q_losses[1][i].copy_(
+                                q_step(q_fns[1], obs, None, targets, _noop_hook,
+                                       qin=qin, step_delta=float(i))
+                            )
+                    main.wait_stream(s1)
+                else:
+                    for q, buf_q in zip(q_fns, self.all_q):
+                        buf_q[i].copy_(forward_only(q, qin).view(-1))
+                    targets = target_chain()
+                    for q, buf_l in zip(q_fns, self.q_losses):
+                        buf_l[i].copy_(
+                            q_step(q, obs, None, targets, _noop_hook, qin=qin,
+                                   step_delta=float(i))
+                        )
                 if i % delay == 0:
                     self.pi_losses[pi_k].copy_(
                         policy_step(algo.policy, q_fns[0], obs, _noop_hook,

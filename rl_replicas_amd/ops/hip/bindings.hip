@@ -217,11 +217,12 @@ std::vector<torch::Tensor> mlp_forward(torch::Tensor x,
   pick_tile(args.batch, max_width, &rows, &maxw);
 
   auto opts = x.options();
-  // per-layer column-split kernels need enough rows to fill the chip
-  // row-wise too; at minibatch scale (DDPG/TD3: 100 rows) the whole-net
-  // <32,256> fused kernel replaces L launches with 1 (weights read
-  // directly from global, wstage mode 2)
-  const bool wide = maxw == 256 && args.batch > 512;
+  // per-layer column-split kernels for wide nets at ANY batch: the
+  // whole-net <32,256> path was measured SLOWER at minibatch scale
+  // (batch 100 -> 4 WGs serializing 3 layers each, 800 vs 1232
+  // env-steps/s on the TD3 bench) — concurrency across independent
+  // per-layer launches beats fewer launches here
+  const bool wide = maxw == 256;
   std::vector<torch::Tensor> outs;  // [final, h0..h_{L-2}]
   torch::Tensor final_out = torch::empty({x.size(0), args.dims[L]}, opts);
   outs.push_back(final_out);
