@@ -161,6 +161,12 @@ class OnPolicyAlgorithm(AlgorithmBase):
         advantages, returns = ops.gae_advantages_and_returns(
             rewards, values, last_values, offsets, dones, self.gamma, self.gae_lambda
         )
+        # replicate-mode DP: ONE fused all-gather of the post-GAE rows;
+        # everything below (normalization, policy/value loops) then runs
+        # on the identical global batch on every rank (parallel/ddp.py)
+        obs, actions, advantages, returns = self._gather_global_batch(
+            obs, actions, advantages, returns
+        )
         advantages = self._normalize_advantages(advantages)
         return {
             "observations": obs,
@@ -168,6 +174,10 @@ class OnPolicyAlgorithm(AlgorithmBase):
             "advantages": advantages.detach(),
             "discounted_returns": returns.detach(),
         }
+
+    def _gather_global_batch(self, obs, actions, advantages, returns):
+        """Hook for replicate-mode DP (identity single-process)."""
+        return obs, actions, advantages, returns
 
     def _normalize_advantages(self, advantages: Tensor) -> Tensor:
         """Global advantage normalization (hook for the DP wrapper)."""

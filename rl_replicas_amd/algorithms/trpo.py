@@ -97,10 +97,15 @@ class TRPO(OnPolicyAlgorithm):
                 self.policy, observations,
                 self.policy.optimizer.hvp_damping_coefficient,
             )
-        # under DP every CG iteration and line-search evaluation must see
-        # the GLOBAL batch (mean over ranks) or replicas diverge
+        # under all-reduce DP every CG iteration and line-search
+        # evaluation must see the GLOBAL batch (mean over ranks) or
+        # replicas diverge; under replicate DP the batch IS global
+        # locally, so no hook (and the captured CG solve stays active)
         reduce_hook = (
-            self._reduce_scalar_mean if getattr(self, "_dp_enabled", False) else None
+            self._reduce_scalar_mean
+            if getattr(self, "_dp_enabled", False)
+            and not getattr(self, "_dp_replicate", False)
+            else None
         )
         self.policy.optimizer.step(
             compute_surrogate_loss, compute_kl_constraint,

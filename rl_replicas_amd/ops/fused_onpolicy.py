@@ -167,10 +167,16 @@ def _graphs_enabled_value(algo) -> bool:
 
 
 def _dp_active(algo) -> bool:
+    """True when the captured loops must SPLIT around eager collectives.
+    Replicate-mode DP runs the whole global batch locally (zero
+    per-iteration collectives), so its loops capture whole like
+    single-process."""
     import os
 
     from rl_replicas_amd.parallel.ddp import distributed_is_active
 
+    if getattr(algo, "_dp_replicate", False):
+        return os.environ.get("RL_REPLICAS_AMD_FORCE_DP_GRAPHS", "0") == "1"
     return (
         distributed_is_active()
         or getattr(algo, "_dp_enabled", False)

@@ -260,10 +260,82 @@ def _worker_td3_sync(rank: int, world: int, tmpdir: str):
     dist.destroy_process_group()
 
 
+def _worker_ppo_sync_allreduce(rank: int, world: int, tmpdir: str):
+    os.environ["RL_REPLICAS_AMD_DP_MODE"] = "allreduce"
+    try:
+        _worker_ppo_sync(rank, world, tmpdir)
+    finally:
+        os.environ.pop("RL_REPLICAS_AMD_DP_MODE", None)
+
+
+def _worker_trpo_sync_allreduce(rank: int, world: int, tmpdir: str):
+    os.environ["RL_REPLICAS_AMD_DP_MODE"] = "allreduce"
+    try:
+        _worker_trpo_sync(rank, world, tmpdir)
+    finally:
+        os.environ.pop("RL_REPLICAS_AMD_DP_MODE", None)
+
+
+def _worker_mode_equivalence(rank: int, world: int, tmpdir: str):
+    """replicate-mode DP (one post-GAE all-gather, local global-batch
+    update) must produce the same update as allreduce-mode DP
+    (per-iteration averaged shard gradients) — both equal the
+    global-batch math, so params agree to fp tolerance."""
+    dist = _init(rank, world, tmpdir)
+    import torch.nn as nn
+
+    from rl_replicas_amd.algorithms import PPO
+    from rl_replicas_amd.experience import Experience
+    from rl_replicas_amd.networks import MLP
+    from rl_replicas_amd.parallel import enable_data_parallel
+    from rl_replicas_amd.policies import GaussianPolicy
+    from rl_replicas_amd.value_function import ValueFunction
+
+    def fixed_experience():
+        rng = np.random.default_rng(500 + rank)  # different shard per rank
+        obs = [[rng.normal(size=5).astype(np.float32) for _ in range(20)]]
+        acts = [[rng.normal(size=2).astype(np.float32) for _ in range(20)]]
+        rews = [[float(rng.normal()) for _ in range(20)]]
+        dones = [[False] * 19 + [True]]
+        last = [rng.normal(size=5).astype(np.float32)]
+        return Experience(obs, acts, rews, last, dones, [0.0], [20])
+
+    class _Null:
+        def record_scalar(self, *a, **k):
+            pass
+
+        def record_phase_ms(self, *a, **k):
+            pass
+
+    def run(mode):
+        torch.manual_seed(7)
+        pnet = MLP([5, 8, 2])
+        log_std = nn.Parameter(-0.5 * torch.ones(2))
+        policy = GaussianPolicy(
+            pnet, torch.optim.Adam(list(pnet.parameters()) + [log_std], lr=3e-4), log_std
+        )
+        vnet = MLP([5, 8, 1])
+        vf = ValueFunction(vnet, torch.optim.Adam(vnet.parameters(), lr=1e-3))
+        model = PPO(policy, vf, None, None, num_policy_gradients=4, num_value_gradients=4)
+        enable_data_parallel(model, mode=mode)
+        model.metrics_manager = _Null()
+        model.current_total_steps = 0
+        model.train(fixed_experience())
+        return [p.detach().clone() for p in list(policy.parameters()) + list(vf.parameters())]
+
+    p_rep = run("replicate")
+    p_red = run("allreduce")
+    for a, b in zip(p_rep, p_red):
+        torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-6)
+    dist.destroy_process_group()
+
+
 @pytest.mark.parametrize(
     "worker",
     [_worker_allreduce, _worker_ppo_sync, _worker_ppo_device_sampler,
-     _worker_trpo_sync, _worker_ddpg_sync, _worker_td3_sync],
+     _worker_trpo_sync, _worker_ddpg_sync, _worker_td3_sync,
+     _worker_ppo_sync_allreduce, _worker_trpo_sync_allreduce,
+     _worker_mode_equivalence],
 )
 def test_two_rank_gloo(worker, tmp_path):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
