@@ -47,6 +47,8 @@ def main() -> None:
     parser.add_argument("--device", default=None, help="cuda / cpu (default: auto)")
     parser.add_argument("--num-envs", type=int, default=20, help="vectorized env instances")
     parser.add_argument("--num-epochs", type=int, default=None, help="override epoch count (for smoke runs)")
+    parser.add_argument("--env-mode", choices=["cpu", "device"], default="cpu",
+                        help="env residency for the synthetic envs (device = GPU-resident rollouts)")
     args = parser.parse_args()
 
     for env_id in args.envs:
@@ -70,6 +72,7 @@ def main() -> None:
                             device=args.device,
                             num_envs=num_envs,
                             num_epochs=args.num_epochs,
+                            env_mode=args.env_mode,
                         )
 
 
