@@ -130,8 +130,11 @@ void check_f32_gpu(const torch::Tensor& t, const char* name) {
 // (4000-row batch -> 125 workgroups); 64 otherwise.
 void pick_tile(int batch, int max_width, int* rows, int* maxw) {
   *maxw = max_width <= 64 ? 64 : 256;
-  *rows = (batch >= 2048 && *maxw == 64) ? 32 : (*maxw == 256 ? 32 : 64);
-  if (*maxw == 64 && batch < 1024) *rows = 16;  // fill more CUs at small batches
+  // narrow nets: 16-row tiles up to 8K rows (4000-row rollout -> 250
+  // WGs; measured +2.4% over 32-row on the PPO bench — latency hiding
+  // beats weight-staging amortization until the chip is well past
+  // full); 32-row beyond
+  *rows = (*maxw == 256) ? 32 : (batch < 8192 ? 16 : 32);
   static int env_rows = []() {
     const char* e = getenv("RL_REPLICAS_AMD_MLP_ROWS");
     return e ? atoi(e) : 0;
