@@ -272,6 +272,7 @@ class _GraphedOffPolicy:
         # every launch on one stream.
         s1 = torch.cuda.Stream() if use_streams else None
         s2 = torch.cuda.Stream() if use_streams else None
+        s3 = torch.cuda.Stream() if (use_streams and twin) else None
 
         def body():
             pi_k = 0
@@ -283,7 +284,7 @@ class _GraphedOffPolicy:
                     minibatch_size, seed, 3 * i, self.ctr,
                 )
 
-                def target_chain():
+                def target_chain(fork_twin: bool):
                     na = forward_only(algo.target_policy, nxt)
                     if twin:
                         na2 = ext.td3_smooth(
@@ -296,8 +297,16 @@ class _GraphedOffPolicy:
                         na2 = na
                     qt_in = torch.cat([nxt, na2], dim=-1).contiguous()
                     if twin:
-                        q1t = forward_only(tq_fns[0], qt_in).view(-1)
-                        q2t = forward_only(tq_fns[1], qt_in).view(-1)
+                        if fork_twin:
+                            cur = torch.cuda.current_stream()
+                            s3.wait_stream(cur)
+                            q1t = forward_only(tq_fns[0], qt_in).view(-1)
+                            with torch.cuda.stream(s3):
+                                q2t = forward_only(tq_fns[1], qt_in).view(-1)
+                            cur.wait_stream(s3)
+                        else:
+                            q1t = forward_only(tq_fns[0], qt_in).view(-1)
+                            q2t = forward_only(tq_fns[1], qt_in).view(-1)
                         return ext.q_target_min2(rew, dn, q1t, q2t, gamma)
                     qt = forward_only(tq_fns[0], qt_in).view(-1)
                     return ext.q_target(rew, dn, qt, gamma)
@@ -313,7 +322,7 @@ class _GraphedOffPolicy:
                                 forward_only(q_fns[1], qin).view(-1)
                             )
                     with torch.cuda.stream(s2):
-                        targets = target_chain()
+                        targets = target_chain(fork_twin=True)
                     main.wait_stream(s1)
                     main.wait_stream(s2)
                     # critic steps in parallel: q1 on main, q2 on s1
@@ -332,7 +341,7 @@ class _GraphedOffPolicy:
                 else:
                     for q, buf_q in zip(q_fns, self.all_q):
                         buf_q[i].copy_(forward_only(q, qin).view(-1))
-                    targets = target_chain()
+                    targets = target_chain(fork_twin=False)
                     for q, buf_l in zip(q_fns, self.q_losses):
                         buf_l[i].copy_(
                             q_step(q, obs, None, targets, _noop_hook, qin=qin,

@@ -26,7 +26,10 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
-                          hipStream_t stream);
+                          hipStream_t stream, int rows);
+// 16-row tiles below 8K rows (measured faster: more WGs hide latency),
+// 32-row beyond — MUST stay consistent with value_loss_partials_blocks
+inline int bwd_fused_rows(int batch) { return batch < 8192 ? 16 : 32; }
 __global__ void mlp_grad_reduce_onepass_f32(ReduceAllArgs a);
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
                                float* out, int batch, int in_d, int out_d,
@@ -301,10 +304,11 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
   // whole-net fused backward for narrow nets when the LDS image fits
   size_t whole_w = 0;
   for (auto& w : weights) whole_w += (size_t)w.size(0) * (w.size(1) + 1);
-  const size_t fused_lds = ((size_t)3 * 32 * 68 + whole_w) * 4;
+  const int brows = bwd_fused_rows(batch);
+  const size_t fused_lds = ((size_t)3 * brows * 68 + whole_w) * 4;
   const bool use_fused_bwd = (maxw == 64) && fused_lds <= 100 * 1024;
   if (use_fused_bwd) {
-    const int fb = (batch + 31) / 32;
+    const int fb = (batch + brows - 1) / brows;
     torch::Tensor ws = torch::empty({(int64_t)fb, grand}, opts);
     std::vector<torch::Tensor> dws(L), dbs(L);
     MLPBwdArgs ba{};
@@ -325,7 +329,8 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     launch_mlp_bwd_fused(ba, x.data_ptr<float>(),
                          grad_out.contiguous().data_ptr<float>(),
                          dx.data_ptr<float>(), ws.data_ptr<float>(), nullptr,
-                         nullptr, fused_lds, fb, (int)compute_bf16, stream);
+                         nullptr, fused_lds, fb, (int)compute_bf16, stream,
+                         brows);
     HIP_OK(hipGetLastError());
 
     ReduceAllArgs ra{};
@@ -575,10 +580,11 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   }
   size_t whole_w = 0;
   for (auto& w : weights) whole_w += (size_t)w.size(0) * (w.size(1) + 1);
-  const size_t fused_lds = ((size_t)3 * 32 * 68 + whole_w) * 4;
+  const int brows = bwd_fused_rows(batch);
+  const size_t fused_lds = ((size_t)3 * brows * 68 + whole_w) * 4;
   TORCH_CHECK(fused_lds <= 100 * 1024, "net too large for fused value backward");
 
-  const int fb = (batch + 31) / 32;
+  const int fb = (batch + brows - 1) / brows;
   torch::Tensor ws = torch::empty({(int64_t)fb, grand}, opts);
   // partials_out: caller-owned slice -> finalize is deferred (the captured
   // value loop batches all iterations' finalizes into one kernel)
@@ -607,7 +613,7 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   launch_mlp_bwd_fused(ba, x.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
                        ws.data_ptr<float>(), returns.data_ptr<float>(),
                        loss_partials.data_ptr<float>(), fused_lds, fb,
-                       (int)compute_bf16, stream);
+                       (int)compute_bf16, stream, brows);
   HIP_OK(hipGetLastError());
 
   ReduceAllArgs ra{};
@@ -991,7 +997,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("final_out"), py::arg("acts"), py::arg("returns"),
         py::arg("compute_bf16") = 0, py::arg("partials_out") = py::none());
   m.def("value_loss_partials_blocks",
-        [](int64_t batch) { return (batch + 31) / 32; },
+        [](int64_t batch) {
+          const int brows = bwd_fused_rows((int)batch);
+          return (batch + brows - 1) / brows;
+        },
         "partials row length used by value_mlp_backward");
   m.def("value_loss_finalize", &value_loss_finalize,
         "batched deferred value-loss finalize (gfx950)");

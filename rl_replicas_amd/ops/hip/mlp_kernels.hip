@@ -709,7 +709,18 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
-                          hipStream_t stream) {
+                          hipStream_t stream, int rows) {
+  if (rows == 16) {
+    if (compute_bf16)
+      hipLaunchKernelGGL((mlp_bwd_fused_f32_t<16, true>), dim3(n_blocks),
+                         dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
+                         mse_returns, loss_partials);
+    else
+      hipLaunchKernelGGL((mlp_bwd_fused_f32_t<16, false>), dim3(n_blocks),
+                         dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
+                         mse_returns, loss_partials);
+    return;
+  }
   if (compute_bf16)
     hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32, true>), dim3(n_blocks),
                        dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
