@@ -284,34 +284,30 @@ class _GraphedOffPolicy:
                     minibatch_size, seed, 3 * i, self.ctr,
                 )
 
-                def target_chain(fork_twin: bool):
+                def smoothed_target_input():
                     na = forward_only(algo.target_policy, nxt)
                     if twin:
-                        na2 = ext.td3_smooth(
+                        na = ext.td3_smooth(
                             na, seed, 3 * i + 1,
                             float(algo.target_noise_scale),
                             float(algo.target_noise_clip), action_limit,
                             self.ctr,
                         )
-                    else:
-                        na2 = na
-                    qt_in = torch.cat([nxt, na2], dim=-1).contiguous()
+                    return torch.cat([nxt, na], dim=-1).contiguous()
+
+                def target_chain():
+                    qt_in = smoothed_target_input()
                     if twin:
-                        if fork_twin:
-                            cur = torch.cuda.current_stream()
-                            s3.wait_stream(cur)
-                            q1t = forward_only(tq_fns[0], qt_in).view(-1)
-                            with torch.cuda.stream(s3):
-                                q2t = forward_only(tq_fns[1], qt_in).view(-1)
-                            cur.wait_stream(s3)
-                        else:
-                            q1t = forward_only(tq_fns[0], qt_in).view(-1)
-                            q2t = forward_only(tq_fns[1], qt_in).view(-1)
+                        q1t = forward_only(tq_fns[0], qt_in).view(-1)
+                        q2t = forward_only(tq_fns[1], qt_in).view(-1)
                         return ext.q_target_min2(rew, dn, q1t, q2t, gamma)
                     qt = forward_only(tq_fns[0], qt_in).view(-1)
                     return ext.q_target(rew, dn, qt, gamma)
 
                 if use_streams:
+                    # every fork originates from (and joins back to) the
+                    # capture-origin stream — nested forks from a side
+                    # stream break hipGraph capture
                     s1.wait_stream(main)
                     s2.wait_stream(main)
                     # main: q1 logging fwd (reference td3.py:230-236)
@@ -322,9 +318,20 @@ class _GraphedOffPolicy:
                                 forward_only(q_fns[1], qin).view(-1)
                             )
                     with torch.cuda.stream(s2):
-                        targets = target_chain(fork_twin=True)
-                    main.wait_stream(s1)
+                        qt_in = smoothed_target_input()
                     main.wait_stream(s2)
+                    if twin:
+                        # q1' on main overlaps q2' on s3
+                        s3.wait_stream(main)
+                        q1t = forward_only(tq_fns[0], qt_in).view(-1)
+                        with torch.cuda.stream(s3):
+                            q2t = forward_only(tq_fns[1], qt_in).view(-1)
+                        main.wait_stream(s3)
+                        targets = ext.q_target_min2(rew, dn, q1t, q2t, gamma)
+                    else:
+                        qt = forward_only(tq_fns[0], qt_in).view(-1)
+                        targets = ext.q_target(rew, dn, qt, gamma)
+                    main.wait_stream(s1)
                     # critic steps in parallel: q1 on main, q2 on s1
                     s1.wait_stream(main)
                     self.q_losses[0][i].copy_(
@@ -341,7 +348,7 @@ class _GraphedOffPolicy:
                 else:
                     for q, buf_q in zip(q_fns, self.all_q):
                         buf_q[i].copy_(forward_only(q, qin).view(-1))
-                    targets = target_chain(fork_twin=False)
+                    targets = target_chain()
                     for q, buf_l in zip(q_fns, self.q_losses):
                         buf_l[i].copy_(
                             q_step(q, obs, None, targets, _noop_hook, qin=qin,
