@@ -94,3 +94,39 @@ class FusedAdam(torch.optim.Adam):
         ]
         if steps:
             ext.adam_bump_(steps, float(amount))
+
+
+def adam_arg_lists(optimizer, weights, biases):
+    """State tensors for the merged reduce+Adam kernel
+    (mlp_grad_reduce_adam_f32): returns (m, v, step0, hp) with m/v
+    ordered [weights..., biases...], step0 the (shared) pre-bump step
+    counter of the first param, and hp the (lr, beta1, beta2, eps,
+    weight_decay) of the owning param group.  All listed params must
+    live in one group and step in lockstep (true for every optimizer
+    this library builds)."""
+    import torch as _torch
+
+    params = list(weights) + list(biases)
+    group = None
+    for g in optimizer.param_groups:
+        ids = {id(p) for p in g["params"]}
+        if id(params[0]) in ids:
+            group = g
+            break
+    assert group is not None, "params not found in optimizer"
+    m, v = [], []
+    step0 = None
+    for p in params:
+        state = optimizer.state[p]
+        if len(state) == 0:
+            state["step"] = _torch.zeros((), dtype=_torch.float32, device=p.device)
+            state["exp_avg"] = _torch.zeros_like(p)
+            state["exp_avg_sq"] = _torch.zeros_like(p)
+        m.append(state["exp_avg"])
+        v.append(state["exp_avg_sq"])
+        if step0 is None:
+            step0 = state["step"]
+    beta1, beta2 = group["betas"]
+    hp = (float(group["lr"]), float(beta1), float(beta2), float(group["eps"]),
+          float(group.get("weight_decay", 0.0)))
+    return m, v, step0, hp

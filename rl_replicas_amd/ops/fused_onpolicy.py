@@ -381,18 +381,29 @@ class _GraphedPPO:
         return {"policy/loss": loss_before, "policy/kl_divergence": approximate_kl}
 
 
-def _value_iter_wide(algo, obs: Tensor, returns: Tensor) -> Tensor:
+def _value_iter_wide(algo, obs: Tensor, returns: Tensor, adam=None) -> Tensor:
     """One fused value MSE step for nets the single-kernel
     value_mlp_backward can't hold (width > 64): fused forward ->
     value-MSE loss kernel -> fused MLP backward.  Same pipeline the
     off-policy critic steps use (fused_offpolicy.q_step), so
     [obs,256,256,1] value nets stay on the kernel path instead of
-    falling back to autograd (round-1 VERDICT weak #6)."""
+    falling back to autograd (round-1 VERDICT weak #6).
+
+    `adam` = (m, v, step0, hp, step_delta): backward's reduction feeds
+    the Adam update in the same kernel (captured-loop fast path) —
+    grads never materialize and the caller must NOT call
+    optimizer.step()."""
     ext = ops._load_extension()
     vf = algo.value_function
     mlp = vf.network
     out, hidden, weights, biases, acts = _forward_saved(mlp, obs)
     dv, scalars = ext.value_mse_loss(out.view(-1), returns)
+    if adam is not None:
+        m, v, step0, hp, delta = adam
+        ext.mlp_backward(dv.view(out.shape), obs, list(weights),
+                         list(biases), list(hidden), out, acts,
+                         ops.compute_bf16(), m, v, step0, *hp, delta)
+        return scalars
     grads = ext.mlp_backward(dv.view(out.shape), obs, list(weights),
                              list(biases), list(hidden), out, acts,
                              ops.compute_bf16())
@@ -449,23 +460,26 @@ class _GraphedValueLoop:
             # steps; same serial partial-sum order per row).
             fb = int(ext.value_loss_partials_blocks(obs0.shape[0]))
             self.partials = torch.zeros(num_iters, fb, device=obs0.device)
+            from rl_replicas_amd.ops.fused_adam import adam_arg_lists
 
-            def iter_deferred(i: int):
-                out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
-                grads = ext.value_mlp_backward(
-                    self.obs, list(weights), list(biases), list(hidden), out,
-                    acts, self.returns, ops.compute_bf16(), self.partials[i],
-                )
-                n = len(weights)
-                for w, dw in zip(weights, grads[1 : 1 + n]):
-                    w.grad = dw
-                for b, db in zip(biases, grads[1 + n : 1 + 2 * n]):
-                    b.grad = db
+            weights0, biases0, _ = _extract_layers(mlp)
+            adam_m, adam_v, adam_step, hp = adam_arg_lists(
+                vf.optimizer, weights0, biases0
+            )
 
             def body():
+                # iteration i: ONE backward kernel + ONE merged
+                # reduce+Adam kernel (mlp_grad_reduce_adam_f32) —
+                # the gradient never round-trips through dw/db tensors
+                weights, biases, acts = _extract_layers(mlp)
                 for i in range(num_iters):
-                    iter_deferred(i)
-                    vf.optimizer.step(step_delta=float(i), do_bump=False)
+                    out, hidden, _, _, _ = _forward_saved(mlp, self.obs)
+                    ext.value_mlp_backward(
+                        self.obs, list(weights), list(biases), list(hidden),
+                        out, acts, self.returns, ops.compute_bf16(),
+                        self.partials[i], adam_m, adam_v, adam_step,
+                        *hp, float(i),
+                    )
                 vf.optimizer.bump_steps(float(num_iters))
                 return ext.value_loss_finalize(self.partials, fb)
 
@@ -485,11 +499,21 @@ class _GraphedValueLoop:
             self.post = _CapturedLoop(lambda: vf.optimizer.step(), state)
         else:
             losses_buf = torch.zeros(num_iters, device=self.obs.device)
+            from rl_replicas_amd.ops.fused_adam import adam_arg_lists
+
+            weights0, biases0, _ = _extract_layers(vf.network)
+            adam_m, adam_v, adam_step, hp = adam_arg_lists(
+                vf.optimizer, weights0, biases0
+            )
 
             def body():
                 for i in range(num_iters):
-                    losses_buf[i].copy_(iter_pre()[0])
-                    vf.optimizer.step(step_delta=float(i), do_bump=False)
+                    losses_buf[i].copy_(
+                        _value_iter_wide(
+                            algo, self.obs, self.returns,
+                            adam=(adam_m, adam_v, adam_step, hp, float(i)),
+                        )[0]
+                    )
                 vf.optimizer.bump_steps(float(num_iters))
                 return losses_buf
 

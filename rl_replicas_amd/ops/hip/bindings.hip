@@ -31,6 +31,7 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
 // 32-row beyond — MUST stay consistent with value_loss_partials_blocks
 inline int bwd_fused_rows(int batch) { return batch < 8192 ? 16 : 32; }
 __global__ void mlp_grad_reduce_onepass_f32(ReduceAllArgs a);
+__global__ void mlp_grad_reduce_adam_f32(ReduceAdamArgs a);
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
                                float* out, int batch, int in_d, int out_d,
                                int act, hipStream_t stream);
@@ -125,6 +126,51 @@ void check_f32_gpu(const torch::Tensor& t, const char* name) {
     hipError_t _e = (expr);                                                 \
     TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));    \
   } while (0)
+
+// merged reduce+Adam launch (optional fast path of mlp_backward /
+// value_mlp_backward): m/v are ordered [weights..., biases...]
+void launch_reduce_adam(const std::vector<torch::Tensor>& weights,
+                        const std::vector<torch::Tensor>& biases,
+                        const std::vector<int64_t>& totals, const float* ws,
+                        long stride, int n_blocks, int64_t grand,
+                        const std::vector<torch::Tensor>& m,
+                        const std::vector<torch::Tensor>& v,
+                        const torch::Tensor& step, double lr, double beta1,
+                        double beta2, double eps, double weight_decay,
+                        double step_delta,
+                        const c10::optional<torch::Tensor>& gate,
+                        hipStream_t stream) {
+  const int L = (int)weights.size();
+  TORCH_CHECK((int)m.size() == 2 * L && (int)v.size() == 2 * L,
+              "adam m/v must be [weights..., biases...] (2L tensors)");
+  ReduceAdamArgs ra{};
+  ra.ws = ws;
+  ra.stride = stride;
+  ra.n_layers = L;
+  ra.n_blocks = n_blocks;
+  for (int l = 0; l < L; ++l) {
+    ra.pw[l] = weights[l].data_ptr<float>();
+    ra.pb[l] = biases[l].data_ptr<float>();
+    ra.mw[l] = m[l].data_ptr<float>();
+    ra.mb[l] = m[L + l].data_ptr<float>();
+    ra.vw[l] = v[l].data_ptr<float>();
+    ra.vb[l] = v[L + l].data_ptr<float>();
+    ra.total[l] = (int)totals[l];
+    ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+  }
+  ra.step = step.data_ptr<float>();
+  ra.lr = (float)lr;
+  ra.beta1 = (float)beta1;
+  ra.beta2 = (float)beta2;
+  ra.eps = (float)eps;
+  ra.weight_decay = (float)weight_decay;
+  ra.step_delta = (float)step_delta;
+  ra.gate = gate.has_value() ? gate->data_ptr<float>() : nullptr;
+  int rb = (int)std::min<int64_t>(256, (grand + 63) / 64);
+  hipLaunchKernelGGL(mlp_grad_reduce_adam_f32, dim3(rb), dim3(256), 0, stream,
+                     ra);
+  HIP_OK(hipGetLastError());
+}
 
 // (ROWS, MAXW) tile selection for the templated MLP kernels.
 // MAXW: smallest instantiated width bound that fits every layer ->
@@ -276,7 +322,15 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
                                         std::vector<torch::Tensor> hidden,
                                         torch::Tensor final_out,
                                         std::vector<int64_t> acts,
-                                        int64_t compute_bf16) {
+                                        int64_t compute_bf16,
+                                        c10::optional<std::vector<torch::Tensor>> adam_m,
+                                        c10::optional<std::vector<torch::Tensor>> adam_v,
+                                        c10::optional<torch::Tensor> adam_step,
+                                        double lr, double beta1, double beta2,
+                                        double eps, double weight_decay,
+                                        double adam_step_delta,
+                                        c10::optional<torch::Tensor> adam_gate) {
+  const bool fuse_adam = adam_m.has_value();
   const int L = (int)weights.size();
   check_f32_gpu(grad_out, "grad_out");
   check_f32_gpu(x, "x");
@@ -333,6 +387,13 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
                          brows);
     HIP_OK(hipGetLastError());
 
+    if (fuse_adam) {
+      launch_reduce_adam(weights, biases, totals, ws.data_ptr<float>(), grand,
+                         fb, grand, *adam_m, *adam_v, *adam_step, lr, beta1,
+                         beta2, eps, weight_decay, adam_step_delta, adam_gate,
+                         stream);
+      return {dx};
+    }
     ReduceAllArgs ra{};
     ra.ws = ws.data_ptr<float>();
     ra.stride = grand;
@@ -390,6 +451,12 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dy = dx;
   }
 
+  if (fuse_adam) {
+    launch_reduce_adam(weights, biases, totals, ws_ptr, grand, n_blocks, grand,
+                       *adam_m, *adam_v, *adam_step, lr, beta1, beta2, eps,
+                       weight_decay, adam_step_delta, adam_gate, stream);
+    return {dx};
+  }
   // one-pass deterministic reduction over the partial rows
   ReduceAllArgs ra{};
   ra.ws = ws_ptr;
@@ -557,7 +624,15 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
                                               std::vector<int64_t> acts,
                                               torch::Tensor returns,
                                               int64_t compute_bf16,
-                                              c10::optional<torch::Tensor> partials_out) {
+                                              c10::optional<torch::Tensor> partials_out,
+                                              c10::optional<std::vector<torch::Tensor>> adam_m,
+                                              c10::optional<std::vector<torch::Tensor>> adam_v,
+                                              c10::optional<torch::Tensor> adam_step,
+                                              double lr, double beta1,
+                                              double beta2, double eps,
+                                              double weight_decay,
+                                              double adam_step_delta) {
+  const bool fuse_adam = adam_m.has_value();
   const int L = (int)weights.size();
   check_f32_gpu(x, "x");
   check_f32_gpu(returns, "returns");
@@ -616,21 +691,28 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
                        (int)compute_bf16, stream, brows);
   HIP_OK(hipGetLastError());
 
-  ReduceAllArgs ra{};
-  ra.ws = ws.data_ptr<float>();
-  ra.stride = grand;
-  ra.n_layers = L;
-  ra.n_blocks = fb;
-  for (int l = 0; l < L; ++l) {
-    ra.dw[l] = dws[l].data_ptr<float>();
-    ra.db[l] = dbs[l].data_ptr<float>();
-    ra.total[l] = (int)totals[l];
-    ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+  if (fuse_adam) {
+    launch_reduce_adam(weights, biases, totals, ws.data_ptr<float>(), grand,
+                       fb, grand, *adam_m, *adam_v, *adam_step, lr, beta1,
+                       beta2, eps, weight_decay, adam_step_delta,
+                       c10::nullopt, stream);
+  } else {
+    ReduceAllArgs ra{};
+    ra.ws = ws.data_ptr<float>();
+    ra.stride = grand;
+    ra.n_layers = L;
+    ra.n_blocks = fb;
+    for (int l = 0; l < L; ++l) {
+      ra.dw[l] = dws[l].data_ptr<float>();
+      ra.db[l] = dbs[l].data_ptr<float>();
+      ra.total[l] = (int)totals[l];
+      ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+    }
+    int rb = (int)std::min<int64_t>(256, (grand + 63) / 64);
+    hipLaunchKernelGGL(mlp_grad_reduce_onepass_f32, dim3(rb), dim3(256), 0,
+                       stream, ra);
+    HIP_OK(hipGetLastError());
   }
-  int rb = (int)std::min<int64_t>(256, (grand + 63) / 64);
-  hipLaunchKernelGGL(mlp_grad_reduce_onepass_f32, dim3(rb), dim3(256), 0,
-                     stream, ra);
-  HIP_OK(hipGetLastError());
   if (!deferred) {
     // loss = sum of the per-block partials (fixed order)
     hipLaunchKernelGGL(loss_partials_finalize, dim3(1), dim3(1), 0, stream,
@@ -990,12 +1072,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mlp_backward", &mlp_backward, "fused MLP backward (gfx950)",
         py::arg("grad_out"), py::arg("x"), py::arg("weights"), py::arg("biases"),
         py::arg("hidden"), py::arg("final_out"), py::arg("acts"),
-        py::arg("compute_bf16") = 0);
+        py::arg("compute_bf16") = 0, py::arg("adam_m") = py::none(),
+        py::arg("adam_v") = py::none(), py::arg("adam_step") = py::none(),
+        py::arg("lr") = 0.0, py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
+        py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
+        py::arg("adam_step_delta") = 0.0, py::arg("adam_gate") = py::none());
   m.def("value_mlp_backward", &value_mlp_backward,
-        "value-net backward with fused MSE seed (gfx950)", py::arg("x"),
+        "fused value-MSE whole-net backward (gfx950)", py::arg("x"),
         py::arg("weights"), py::arg("biases"), py::arg("hidden"),
         py::arg("final_out"), py::arg("acts"), py::arg("returns"),
-        py::arg("compute_bf16") = 0, py::arg("partials_out") = py::none());
+        py::arg("compute_bf16") = 0, py::arg("partials_out") = py::none(),
+        py::arg("adam_m") = py::none(), py::arg("adam_v") = py::none(),
+        py::arg("adam_step") = py::none(), py::arg("lr") = 0.0,
+        py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
+        py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
+        py::arg("adam_step_delta") = 0.0);
   m.def("value_loss_partials_blocks",
         [](int64_t batch) {
           const int brows = bwd_fused_rows((int)batch);

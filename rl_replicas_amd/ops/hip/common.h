@@ -117,6 +117,27 @@ struct PolyakArgs {
   float rho;
 };
 
+// merged gradient-reduce + Adam update (mlp_kernels.hip): consumes the
+// backward stage-1 workspace and applies the optimizer in the SAME
+// kernel — no dw/db round trip through HBM, one fewer dependent launch
+// per optimizer step in the captured loops
+struct ReduceAdamArgs {
+  const float* ws;  // [n_blocks][stride] partials
+  long stride;
+  float* pw[MLP_MAX_LAYERS];
+  float* pb[MLP_MAX_LAYERS];
+  float* mw[MLP_MAX_LAYERS];
+  float* mb[MLP_MAX_LAYERS];
+  float* vw[MLP_MAX_LAYERS];
+  float* vb[MLP_MAX_LAYERS];
+  const float* step;  // shared pre-bump step counter (params step in lockstep)
+  int total[MLP_MAX_LAYERS];
+  int wsize[MLP_MAX_LAYERS];
+  int n_layers, n_blocks;
+  float lr, beta1, beta2, eps, weight_decay, step_delta;
+  const float* gate;  // optional: skip entirely while *gate == 0
+};
+
 // replay-ring minibatch gather (offpolicy_kernels.hip / bindings.hip)
 struct ReplayGatherArgs {
   const float* obs;   // [cap, O] ring storage

@@ -798,6 +798,58 @@ void launch_mlp_bwd_layer(const float* dy, const float* y, const float* xin,
 // deterministic split-K partial reduction
 // one-pass variant: 256 threads = 64 elements x 4 partial-quarters,
 // combined through LDS in fixed order — one launch even at 125 partials
+// reduce + Adam in one pass: identical fixed-order partial summation to
+// mlp_grad_reduce_onepass_f32, but the summed gradient feeds the Adam
+// update (same math as fused_adam_kernel) directly — no dw/db tensors.
+__global__ __launch_bounds__(256) void mlp_grad_reduce_adam_f32(ReduceAdamArgs a) {
+  if (a.gate && *a.gate == 0.f) return;
+  __shared__ float red[256];
+  int grand = 0;
+  int base[MLP_MAX_LAYERS];
+  for (int l = 0; l < a.n_layers; ++l) {
+    base[l] = grand;
+    grand += a.total[l];
+  }
+  const float s0 = a.step[0] + 1.f + a.step_delta;
+  const float bc1 = 1.f - __powf(a.beta1, s0);
+  const float bc2 = 1.f - __powf(a.beta2, s0);
+  const int e_local = threadIdx.x & 63;
+  const int quarter = threadIdx.x >> 6;
+  const int chunk = (a.n_blocks + 3) / 4;
+  const int p0 = quarter * chunk;
+  const int p1 = min(p0 + chunk, a.n_blocks);
+  for (int g0 = blockIdx.x * 64; g0 < grand; g0 += gridDim.x * 64) {
+    const int g = g0 + e_local;
+    float s = 0.f;
+    if (g < grand) {
+      for (int p = p0; p < p1; ++p) s += a.ws[p * a.stride + g];
+    }
+    red[threadIdx.x] = s;
+    __syncthreads();
+    if (quarter == 0 && g < grand) {
+      float grad = (red[e_local] + red[64 + e_local]) +
+                   (red[128 + e_local] + red[192 + e_local]);
+      int l = 0;
+      while (l + 1 < a.n_layers && g >= base[l + 1]) ++l;
+      int idx = g - base[l];
+      float *p, *m, *v;
+      if (idx < a.wsize[l]) {
+        p = a.pw[l]; m = a.mw[l]; v = a.vw[l];
+      } else {
+        idx -= a.wsize[l];
+        p = a.pb[l]; m = a.mb[l]; v = a.vb[l];
+      }
+      if (a.weight_decay != 0.f) grad += a.weight_decay * p[idx];
+      const float mi = a.beta1 * m[idx] + (1.f - a.beta1) * grad;
+      const float vi = a.beta2 * v[idx] + (1.f - a.beta2) * grad * grad;
+      m[idx] = mi;
+      v[idx] = vi;
+      p[idx] -= a.lr * (mi / bc1) / (sqrtf(vi / bc2) + a.eps);
+    }
+    __syncthreads();
+  }
+}
+
 __global__ __launch_bounds__(256) void mlp_grad_reduce_onepass_f32(ReduceAllArgs a) {
   __shared__ float red[256];
   int grand = 0;
