@@ -257,8 +257,8 @@ class _GraphedPPO:
         self.old_logp = old_logp0.clone()
         clip = float(algo.clip_range)
 
-        def body_pre():
-            out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
+        def loss_bwd_from(out, hidden, weights, biases, acts):
+            """loss + gradients from an already-computed saved forward."""
             if kind == "gaussian":
                 dmean, dlog_std, scalars = ext.gaussian_policy_loss(
                     out, self.actions, self.old_logp, self.adv,
@@ -284,12 +284,18 @@ class _GraphedPPO:
                 param.grad = grad
             return scalars
 
-        def kl_eval():
-            new_out = _forward_only(mlp, self.obs)
+        def body_pre():
+            out, hidden, weights, biases, acts = _forward_saved(mlp, self.obs)
+            return loss_bwd_from(out, hidden, weights, biases, acts)
+
+        def kl_from_out(out):
             if kind == "gaussian":
-                return ext.gaussian_kl(new_out, self.actions,
+                return ext.gaussian_kl(out, self.actions,
                                        policy.log_std.data, self.old_logp)
-            return ext.categorical_kl(new_out, self.actions, self.old_logp)
+            return ext.categorical_kl(out, self.actions, self.old_logp)
+
+        def kl_eval():
+            return kl_from_out(_forward_only(mlp, self.obs))
 
         state = [p.data for p in policy.parameters()]
         state += _ensure_adam_state(policy.optimizer)
@@ -317,19 +323,36 @@ class _GraphedPPO:
                         self.gate.fill_(1.0)
                         self.iters_done.zero_()
                     for i in range(start, start + count):
-                        scalars = body_pre()
+                        # iteration i's saved forward evaluates the policy
+                        # at params_i — which is EXACTLY the network output
+                        # iteration i-1's post-step KL test needs, so one
+                        # forward serves both (the reference recomputes it,
+                        # ppo.py:176-181)
+                        out, hidden, weights, biases, acts = _forward_saved(
+                            mlp, self.obs
+                        )
+                        if i > 0:
+                            kl = kl_from_out(out)
+                            # one-kernel bookkeeping: while active, record
+                            # the KL (the reported value is the stop-
+                            # triggering one) and the iteration count, then
+                            # close the gate if KL > thr — the step below
+                            # is then skipped, like the reference's break
+                            ext.ppo_gate_update_(self.gate, kl, self.kl_final,
+                                                 self.iters_done, thr)
+                        scalars = loss_bwd_from(out, hidden, weights, biases,
+                                                acts)
                         if i == 0:
                             self.loss0.copy_(scalars[:1])
                         policy.optimizer.step(gate=self.gate)
-                        kl = kl_eval()
-                        # one-kernel bookkeeping: while active, record the
-                        # KL (the reported value is the stop-triggering
-                        # one, reference ppo.py:176-181) and the iteration
-                        # count, then close the gate if KL > thr
-                        ext.ppo_gate_update_(self.gate, kl, self.kl_final,
-                                             self.iters_done, thr)
 
                 return chunk
+
+            def epilogue():
+                # the final executed iteration's KL is still pending
+                kl = kl_eval()
+                ext.ppo_gate_update_(self.gate, kl, self.kl_final,
+                                     self.iters_done, thr)
 
             # chunked capture: the KL early stop is common in steady
             # state, and a gate-frozen iteration still executes its
@@ -342,6 +365,7 @@ class _GraphedPPO:
             for start in range(0, num_iters, CHUNK):
                 count = min(CHUNK, num_iters - start)
                 self.chunks.append(_CapturedLoop(make_chunk(start, count), state))
+            self.epilogue = _CapturedLoop(epilogue, state)
 
     def run(self, algo, obs, actions, advantages, old_logp) -> Dict[str, float]:
         self.obs.copy_(obs)
@@ -354,6 +378,9 @@ class _GraphedPPO:
                 chunk.replay()
                 if float(self.gate[0]) == 0.0:
                     break
+            # flush the last executed iteration's pending KL (no-op if
+            # the gate already closed)
+            self.epilogue.replay()
             iters = int(self.iters_done)
             if iters < algo.num_policy_gradients:
                 logger.info(
