@@ -138,10 +138,14 @@ def policy_step(policy, q_function, observations: Tensor, all_reduce_hook,
     q_out, q_hidden, q_w, q_b, q_acts = _fwd_saved(qm, qin)
 
     # d(-mean(q))/dq = -1/B ; propagate to the Q input, take the action
-    # columns, then backprop through the actor.  Q's weight grads are
-    # computed but never applied (the reference freezes the critic).
+    # columns, then backprop through the actor.  The critic is frozen
+    # (reference semantics), so only the dgrad chain runs — its weight
+    # gradients are never computed (input_grad_only).
     dq = torch.full_like(q_out, -1.0 / B)
-    q_grads = _backward(qm, qin, dq, q_hidden, q_out, q_w, q_b, q_acts)
+    q_grads = ext.mlp_backward(
+        dq.contiguous(), qin, list(q_w), list(q_b), list(q_hidden), q_out,
+        q_acts, ops.compute_bf16(), input_grad_only=True,
+    )
     d_qin = q_grads[0]
     d_act = d_qin[:, observations.shape[1] :].contiguous()
 

@@ -39,6 +39,9 @@ void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
                          const float* W, float* dx, float* ws, long ws_stride,
                          int batch, int out_d, int in_d, int act,
                          hipStream_t stream);
+void launch_mlp_dgrad_wide(const float* dy, const float* y, const float* W,
+                           float* dx, int batch, int out_d, int in_d, int act,
+                           hipStream_t stream);
 void launch_gaussian_loss(const float* mean, const float* actions,
                           const float* old_logp, const float* adv,
                           const float* log_std, float* dmean, float* c_buf,
@@ -329,7 +332,8 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
                                         double lr, double beta1, double beta2,
                                         double eps, double weight_decay,
                                         double adam_step_delta,
-                                        c10::optional<torch::Tensor> adam_gate) {
+                                        c10::optional<torch::Tensor> adam_gate,
+                                        bool input_grad_only) {
   const bool fuse_adam = adam_m.has_value();
   const int L = (int)weights.size();
   check_f32_gpu(grad_out, "grad_out");
@@ -387,6 +391,7 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
                          brows);
     HIP_OK(hipGetLastError());
 
+    if (input_grad_only) return {dx};
     if (fuse_adam) {
       launch_reduce_adam(weights, biases, totals, ws.data_ptr<float>(), grand,
                          fb, grand, *adam_m, *adam_v, *adam_step, lr, beta1,
@@ -431,11 +436,20 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dbs[l] = torch::empty({out_d}, opts);
     dx = torch::empty({batch, in_d}, opts);
     if (maxw == 256) {
-      // wide layer: 2D-grid dgrad + wgrad kernels
-      launch_mlp_bwd_wide(dy.data_ptr<float>(), y.data_ptr<float>(),
-                          xin.data_ptr<float>(), weights[l].data_ptr<float>(),
-                          dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
-                          batch, out_d, in_d, (int)acts[l], stream);
+      if (input_grad_only) {
+        // dgrad chain only (e.g. actor step through a frozen critic:
+        // the critic's weight grads would be discarded)
+        launch_mlp_dgrad_wide(dy.data_ptr<float>(), y.data_ptr<float>(),
+                              weights[l].data_ptr<float>(),
+                              dx.data_ptr<float>(), batch, out_d, in_d,
+                              (int)acts[l], stream);
+      } else {
+        // wide layer: 2D-grid dgrad + wgrad kernels
+        launch_mlp_bwd_wide(dy.data_ptr<float>(), y.data_ptr<float>(),
+                            xin.data_ptr<float>(), weights[l].data_ptr<float>(),
+                            dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
+                            batch, out_d, in_d, (int)acts[l], stream);
+      }
     } else {
       int wmode;
       size_t lds_bytes;
@@ -451,6 +465,7 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
     dy = dx;
   }
 
+  if (input_grad_only) return {dx};
   if (fuse_adam) {
     launch_reduce_adam(weights, biases, totals, ws_ptr, grand, n_blocks, grand,
                        *adam_m, *adam_v, *adam_step, lr, beta1, beta2, eps,
@@ -1076,7 +1091,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("adam_v") = py::none(), py::arg("adam_step") = py::none(),
         py::arg("lr") = 0.0, py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
         py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
-        py::arg("adam_step_delta") = 0.0, py::arg("adam_gate") = py::none());
+        py::arg("adam_step_delta") = 0.0, py::arg("adam_gate") = py::none(),
+        py::arg("input_grad_only") = false);
   m.def("value_mlp_backward", &value_mlp_backward,
         "fused value-MSE whole-net backward (gfx950)", py::arg("x"),
         py::arg("weights"), py::arg("biases"), py::arg("hidden"),

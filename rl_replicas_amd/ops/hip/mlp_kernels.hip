@@ -757,6 +757,19 @@ void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
                      out_d, in_d, act);
 }
 
+// dgrad chain only (input gradient without weight-grad partials) — the
+// actor step's path through the frozen critic
+void launch_mlp_dgrad_wide(const float* dy, const float* y, const float* W,
+                           float* dx, int batch, int out_d, int in_d, int act,
+                           hipStream_t stream) {
+  constexpr int ROWS = 32;
+  const int rb = (batch + ROWS - 1) / ROWS;
+  size_t lds1 = (size_t)ROWS * (256 + 4) * 4;
+  hipLaunchKernelGGL((mlp_dgrad_wide_f32<ROWS>), dim3(rb, (in_d + 63) / 64),
+                     dim3(256), lds1, stream, dy, y, W, dx, batch, out_d, in_d,
+                     act);
+}
+
 void launch_mlp_fwd(const MLPArgs& args, const float* x, int save_hidden,
                     int rows, int maxw, int n_blocks, int wstage_mode,
                     size_t lds_bytes, int compute_bf16, hipStream_t stream) {
