@@ -581,18 +581,17 @@ __global__ __launch_bounds__(256) void mlp_layer_fwd_wide_f32(
   }
 }
 
-template <int ROWS>
-__global__ __launch_bounds__(256) void mlp_dgrad_wide_f32(
-    const float* __restrict__ dy, const float* __restrict__ y,
-    const float* __restrict__ W, float* __restrict__ dx, int batch, int out_d,
-    int in_d, int act) {
+DEV_INLINE void dgrad_wide_body(
+    float* __restrict__ smem, const float* __restrict__ dy,
+    const float* __restrict__ y, const float* __restrict__ W,
+    float* __restrict__ dx, int batch, int out_d, int in_d, int act,
+    int bx, int by, int ROWS) {
   constexpr int LDSW = 256 + 4;
-  extern __shared__ float smem[];  // dz tile [ROWS][LDSW]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int row0 = blockIdx.x * ROWS;
-  const int jt = blockIdx.y * 64 + wave * 16;  // over in_d
+  const int row0 = bx * ROWS;
+  const int jt = by * 64 + wave * 16;  // over in_d
 
   for (int idx = tid; idx < ROWS * out_d; idx += 256) {
     int r = idx / out_d, c = idx % out_d;
@@ -640,23 +639,32 @@ __global__ __launch_bounds__(256) void mlp_dgrad_wide_f32(
   }
 }
 
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_dgrad_wide_f32(
+    const float* __restrict__ dy, const float* __restrict__ y,
+    const float* __restrict__ W, float* __restrict__ dx, int batch, int out_d,
+    int in_d, int act) {
+  extern __shared__ float smem[];
+  dgrad_wide_body(smem, dy, y, W, dx, batch, out_d, in_d, act, blockIdx.x,
+                  blockIdx.y, ROWS);
+}
+
 // wgrad + bias partials, columns of dW's out dimension split over
 // blockIdx.y (disjoint writes into the same per-row-block partial row)
-template <int ROWS>
-__global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
-    const float* __restrict__ dy, const float* __restrict__ y,
-    const float* __restrict__ xin, float* __restrict__ workspace,
-    long ws_stride, int batch, int out_d, int in_d, int act) {
+DEV_INLINE void wgrad_wide_body(
+    float* __restrict__ smem, const float* __restrict__ dy,
+    const float* __restrict__ y, const float* __restrict__ xin,
+    float* __restrict__ workspace, long ws_stride, int batch, int out_d,
+    int in_d, int act, int bx, int by, int ROWS) {
   constexpr int LDSW = 256 + 4;
   constexpr int DZW = 64 + 4;
-  extern __shared__ float smem[];  // xt [ROWS][LDSW] + dz slice [ROWS][DZW]
   float* xt = smem;
   float* dz = smem + ROWS * LDSW;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int row0 = blockIdx.x * ROWS;
-  const int og0 = blockIdx.y * 64;  // out-col group
+  const int row0 = bx * ROWS;
+  const int og0 = by * 64;  // out-col group
   float* wsp = workspace + (long)blockIdx.x * ws_stride;
 
   load_tile<LDSW>(xin, xt, row0, batch, in_d, tid, ROWS);
@@ -700,6 +708,38 @@ __global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
     #pragma unroll 4
     for (int r = 0; r < ROWS; ++r) s += dz[r * DZW + c];
     wsp[(long)out_d * in_d + og0 + c] = s;
+  }
+}
+
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_wgrad_wide_f32(
+    const float* __restrict__ dy, const float* __restrict__ y,
+    const float* __restrict__ xin, float* __restrict__ workspace,
+    long ws_stride, int batch, int out_d, int in_d, int act) {
+  extern __shared__ float smem[];
+  wgrad_wide_body(smem, dy, y, xin, workspace, ws_stride, batch, out_d, in_d,
+                  act, blockIdx.x, blockIdx.y, ROWS);
+}
+
+// dgrad and wgrad of one layer are independent given dZ: one launch with
+// blockIdx.z selecting the family doubles the resident block pool (the
+// per-family grids are 16ish blocks at minibatch scale — far below the
+// 256-CU chip) instead of serializing two launches on the stream
+template <int ROWS>
+__global__ __launch_bounds__(256) void mlp_bwd_wide_both_f32(
+    const float* __restrict__ dy, const float* __restrict__ y,
+    const float* __restrict__ xin, const float* __restrict__ W,
+    float* __restrict__ dx, float* __restrict__ workspace, long ws_stride,
+    int batch, int out_d, int in_d, int act) {
+  extern __shared__ float smem[];
+  if (blockIdx.z == 0) {
+    if ((int)blockIdx.y * 64 < in_d)
+      dgrad_wide_body(smem, dy, y, W, dx, batch, out_d, in_d, act, blockIdx.x,
+                      blockIdx.y, ROWS);
+  } else {
+    if ((int)blockIdx.y * 64 < out_d)
+      wgrad_wide_body(smem, dy, y, xin, workspace, ws_stride, batch, out_d,
+                      in_d, act, blockIdx.x, blockIdx.y, ROWS);
   }
 }
 
@@ -747,13 +787,11 @@ void launch_mlp_bwd_wide(const float* dy, const float* y, const float* xin,
                          hipStream_t stream) {
   constexpr int ROWS = 32;
   const int rb = (batch + ROWS - 1) / ROWS;
-  size_t lds1 = (size_t)ROWS * (256 + 4) * 4;
-  hipLaunchKernelGGL((mlp_dgrad_wide_f32<ROWS>), dim3(rb, (in_d + 63) / 64),
-                     dim3(256), lds1, stream, dy, y, W, dx, batch, out_d, in_d,
-                     act);
-  size_t lds2 = (size_t)ROWS * (256 + 4 + 64 + 4) * 4;
-  hipLaunchKernelGGL((mlp_wgrad_wide_f32<ROWS>), dim3(rb, (out_d + 63) / 64),
-                     dim3(256), lds2, stream, dy, y, xin, ws, ws_stride, batch,
+  const int it = (in_d + 63) / 64, ot = (out_d + 63) / 64;
+  const int yt = it > ot ? it : ot;
+  size_t lds = (size_t)ROWS * (256 + 4 + 64 + 4) * 4;  // max of both bodies
+  hipLaunchKernelGGL((mlp_bwd_wide_both_f32<ROWS>), dim3(rb, yt, 2), dim3(256),
+                     lds, stream, dy, y, xin, W, dx, ws, ws_stride, batch,
                      out_d, in_d, act);
 }
 
