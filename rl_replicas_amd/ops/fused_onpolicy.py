@@ -494,19 +494,32 @@ class _GraphedValueLoop:
                 vf.optimizer, weights0, biases0
             )
 
+            fp32 = ops.compute_bf16() == 0
+            dummy = torch.empty(0, device=obs0.device)
+
             def body():
-                # iteration i: ONE backward kernel + ONE merged
-                # reduce+Adam kernel (mlp_grad_reduce_adam_f32) —
-                # the gradient never round-trips through dw/db tensors
+                # iteration i (fp32): ONE fwd+MSE+backward kernel with
+                # LDS-resident activations (fwd_in_kernel) + ONE merged
+                # reduce+Adam kernel — two dependent launches per value
+                # iteration, nothing round-trips through HBM except the
+                # gradient partials
                 weights, biases, acts = _extract_layers(mlp)
                 for i in range(num_iters):
-                    out, hidden, _, _, _ = _forward_saved(mlp, self.obs)
-                    ext.value_mlp_backward(
-                        self.obs, list(weights), list(biases), list(hidden),
-                        out, acts, self.returns, ops.compute_bf16(),
-                        self.partials[i], adam_m, adam_v, adam_step,
-                        *hp, float(i),
-                    )
+                    if fp32:
+                        ext.value_mlp_backward(
+                            self.obs, list(weights), list(biases), [],
+                            dummy, acts, self.returns, 0,
+                            self.partials[i], adam_m, adam_v, adam_step,
+                            *hp, float(i), True,
+                        )
+                    else:
+                        out, hidden, _, _, _ = _forward_saved(mlp, self.obs)
+                        ext.value_mlp_backward(
+                            self.obs, list(weights), list(biases), list(hidden),
+                            out, acts, self.returns, ops.compute_bf16(),
+                            self.partials[i], adam_m, adam_v, adam_step,
+                            *hp, float(i),
+                        )
                 vf.optimizer.bump_steps(float(num_iters))
                 return ext.value_loss_finalize(self.partials, fb)
 

@@ -26,7 +26,7 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
-                          hipStream_t stream, int rows);
+                          hipStream_t stream, int rows, int do_fwd = 0);
 // 16-row tiles below 8K rows (measured faster: more WGs hide latency),
 // 32-row beyond — MUST stay consistent with value_loss_partials_blocks
 inline int bwd_fused_rows(int batch) { return batch < 8192 ? 16 : 32; }
@@ -646,8 +646,11 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
                                               double lr, double beta1,
                                               double beta2, double eps,
                                               double weight_decay,
-                                              double adam_step_delta) {
+                                              double adam_step_delta,
+                                              bool fwd_in_kernel) {
   const bool fuse_adam = adam_m.has_value();
+  TORCH_CHECK(!fwd_in_kernel || compute_bf16 == 0,
+              "fwd_in_kernel is fp32-only");
   const int L = (int)weights.size();
   check_f32_gpu(x, "x");
   check_f32_gpu(returns, "returns");
@@ -671,7 +674,8 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   size_t whole_w = 0;
   for (auto& w : weights) whole_w += (size_t)w.size(0) * (w.size(1) + 1);
   const int brows = bwd_fused_rows(batch);
-  const size_t fused_lds = ((size_t)3 * brows * 68 + whole_w) * 4;
+  const size_t act_tiles = (size_t)(fwd_in_kernel ? 3 + L : 3);
+  const size_t fused_lds = (act_tiles * brows * 68 + whole_w) * 4;
   TORCH_CHECK(fused_lds <= 100 * 1024, "net too large for fused value backward");
 
   const int fb = (batch + brows - 1) / brows;
@@ -692,7 +696,12 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   ba.dims[0] = (int)x.size(1);
   for (int l = 0; l < L; ++l) {
     ba.w[l] = weights[l].data_ptr<float>();
-    ba.h[l] = (l == L - 1 ? final_out : hidden[l]).data_ptr<float>();
+    // fwd_in_kernel: activations never exist in HBM — the kernel
+    // computes them into LDS (hidden/final_out are ignored)
+    ba.h[l] = fwd_in_kernel
+                  ? nullptr
+                  : (l == L - 1 ? final_out : hidden[l]).data_ptr<float>();
+    ba.b[l] = biases[l].data_ptr<float>();
     ba.dims[l + 1] = (int)weights[l].size(0);
     ba.acts[l] = (int)acts[l];
     ba.layer_off[l] = (int)layer_off[l];
@@ -703,7 +712,7 @@ std::vector<torch::Tensor> value_mlp_backward(torch::Tensor x,
   launch_mlp_bwd_fused(ba, x.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
                        ws.data_ptr<float>(), returns.data_ptr<float>(),
                        loss_partials.data_ptr<float>(), fused_lds, fb,
-                       (int)compute_bf16, stream, brows);
+                       (int)compute_bf16, stream, brows, fwd_in_kernel ? 1 : 0);
   HIP_OK(hipGetLastError());
 
   if (fuse_adam) {
@@ -1102,7 +1111,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("adam_step") = py::none(), py::arg("lr") = 0.0,
         py::arg("beta1") = 0.9, py::arg("beta2") = 0.999,
         py::arg("eps") = 1e-8, py::arg("weight_decay") = 0.0,
-        py::arg("adam_step_delta") = 0.0);
+        py::arg("adam_step_delta") = 0.0, py::arg("fwd_in_kernel") = false);
   m.def("value_loss_partials_blocks",
         [](int64_t batch) {
           const int brows = bwd_fused_rows((int)batch);

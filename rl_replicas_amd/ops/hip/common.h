@@ -66,6 +66,7 @@ struct MLPArgs {
 struct MLPBwdArgs {
   const float* w[MLP_MAX_LAYERS];
   const float* h[MLP_MAX_LAYERS];  // post-activations; h[L-1] = final out
+  const float* b[MLP_MAX_LAYERS];  // biases (DO_FWD mode only)
   int dims[MLP_MAX_LAYERS + 1];
   int acts[MLP_MAX_LAYERS];
   int n_layers;

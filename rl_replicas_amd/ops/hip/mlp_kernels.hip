@@ -332,7 +332,11 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
 // from the value-MSE gradient 2*(v - ret)/B (final activation must be
 // identity, out_d == 1 — the value-function case, ppo.py:283-287) and
 // the per-block loss partial sum((v-ret)^2)/B goes to loss_partials.
-template <int ROWS, bool BF16 = false>
+// DO_FWD: the forward pass runs INSIDE this kernel (value-loop fast
+// path): x is staged once, every layer's activations land in LDS
+// (hlds) and never touch HBM; the MSE seed and the backward read them
+// from LDS.  fp32 only; requires mse_returns.
+template <int ROWS, bool BF16 = false, bool DO_FWD = false>
 __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
     MLPBwdArgs args, const float* __restrict__ x, const float* __restrict__ dy,
     float* __restrict__ dx_out, float* __restrict__ workspace,
@@ -357,12 +361,51 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
 
   // whole-net padded W image + layer offsets
   int woffs[MLP_MAX_LAYERS];
+  int wtotal = 0;
   {
     int off = 0;
     for (int l = 0; l < L; ++l) {
       woffs[l] = off;
       stage_weights_block(args.w[l], wlds + off, args.dims[l + 1], args.dims[l], tid);
       off += args.dims[l + 1] * (args.dims[l] + 1);
+    }
+    wtotal = off;
+  }
+  // DO_FWD: per-layer activations live here, [L][ROWS][LDSW]
+  float* const hlds = wlds + wtotal;
+  if constexpr (DO_FWD) {
+    const int fi = lane & 15;
+    load_tile<LDSW>(x, xt, row0, args.batch, args.dims[0], tid, ROWS);
+    __syncthreads();  // weights + x staged
+    const float* fin = xt;
+    for (int l = 0; l < L; ++l) {
+      const int in_d = args.dims[l];
+      const int out_d = args.dims[l + 1];
+      const int wrow = in_d + 1;
+      const float* wl = wlds + woffs[l];
+      float* hl = hlds + l * ROWS * LDSW;
+      const int fk = lane >> 4;
+      for (int jt = jt0; jt < out_d; jt += 16 * JT_STRIDE) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+        const int j = jt + fi;
+        const bool jok = j < out_d;
+        for (int k0 = 0; k0 < in_d; k0 += 4) {
+          const int kk = k0 + fk;
+          float a = (kk < in_d) ? fin[(wr0 + fi) * LDSW + kk] : 0.f;
+          float bv = (jok && kk < in_d) ? wl[j * wrow + kk] : 0.f;
+          acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
+        }
+        if (jok) {
+          const float bias = args.b[l][j];
+          #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int rr = wr0 + (lane >> 4) * 4 + r;
+            hl[rr * LDSW + j] = act_apply(args.acts[l], acc[r] + bias);
+          }
+        }
+      }
+      __syncthreads();
+      fin = hl;
     }
   }
   // seed dZ for the last layer: dY * act'(final out), or the fused
@@ -371,6 +414,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
     const int od = args.dims[L];
     const int act = args.acts[L - 1];
     const float* yl = args.h[L - 1];
+    const float* yl_lds = hlds + (L - 1) * ROWS * LDSW;
     const float inv_b = 2.f / (float)args.batch;
     float loss_acc = 0.f;
     for (int idx = tid; idx < ROWS * od; idx += 256) {
@@ -378,13 +422,18 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
       int row = row0 + r;
       float v = 0.f;
       if (row < args.batch) {
-        long g = (long)row * od + c;
+        float yv;
+        if constexpr (DO_FWD) {
+          yv = yl_lds[r * LDSW + c];
+        } else {
+          yv = yl[(long)row * od + c];
+        }
         if (mse_returns) {
-          const float diff = yl[g] - mse_returns[row];
+          const float diff = yv - mse_returns[row];
           v = diff * inv_b;
           loss_acc += diff * diff;
         } else {
-          v = dy[g] * act_grad_from_y(act, yl[g]);
+          v = dy[(long)row * od + c] * act_grad_from_y(act, yv);
         }
       }
       dza[r * LDSW + c] = v;
@@ -414,9 +463,16 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
     const int wrow = in_d + 1;
     const float* wl = wlds + woffs[l];
 
-    // stage X_l (input activations of layer l; post-act of layer l-1)
-    load_tile<LDSW>(l == 0 ? x : args.h[l - 1], xt, row0, args.batch, in_d, tid, ROWS);
-    __syncthreads();
+    // X_l (input activations of layer l; post-act of layer l-1):
+    // DO_FWD reads them straight from the LDS-resident forward
+    const float* xl;
+    if constexpr (DO_FWD) {
+      xl = (l == 0) ? xt : hlds + (l - 1) * ROWS * LDSW;
+    } else {
+      load_tile<LDSW>(l == 0 ? x : args.h[l - 1], xt, row0, args.batch, in_d, tid, ROWS);
+      __syncthreads();
+      xl = xt;
+    }
 
     // ---- wgrad partials: dW[i][j] = sum_r dZ[r][i] X[r][j] ----
     const int n_it = (out_d + 15) / 16;
@@ -432,14 +488,14 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
             for (int e = 0; e < 8; ++e) {
               const int kk = k0 + k * 8 + e;
               af[e] = f32_to_bf16((kk < ROWS && ii < out_d) ? dz_cur[kk * LDSW + ii] : 0.f);
-              bf[e] = f32_to_bf16((kk < ROWS && jt + i < in_d) ? xt[kk * LDSW + jt + i] : 0.f);
+              bf[e] = f32_to_bf16((kk < ROWS && jt + i < in_d) ? xl[kk * LDSW + jt + i] : 0.f);
             }
             acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
           }
         } else {
           for (int k0 = 0; k0 < ROWS; k0 += 4) {
             float a = (ii < out_d) ? dz_cur[(k0 + k) * LDSW + ii] : 0.f;
-            float bv = (jt + i < in_d) ? xt[(k0 + k) * LDSW + jt + i] : 0.f;
+            float bv = (jt + i < in_d) ? xl[(k0 + k) * LDSW + jt + i] : 0.f;
             acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bv, acc, 0, 0, 0);
           }
         }
@@ -492,7 +548,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
         for (int r = 0; r < 4; ++r) {
           const int row = wr0 + (lane >> 4) * 4 + r;
           if (l > 0) {
-            const float g = act_grad_from_y(prev_act, xt[row * LDSW + j]);
+            const float g = act_grad_from_y(prev_act, xl[row * LDSW + j]);
             dz_nxt[row * LDSW + j] = acc[r] * g;
           } else {
             const int grow = row0 + row;
@@ -749,7 +805,18 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
-                          hipStream_t stream, int rows) {
+                          hipStream_t stream, int rows, int do_fwd) {
+  if (do_fwd) {  // fp32 only (host gates)
+    if (rows == 16)
+      hipLaunchKernelGGL((mlp_bwd_fused_f32_t<16, false, true>), dim3(n_blocks),
+                         dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
+                         mse_returns, loss_partials);
+    else
+      hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32, false, true>), dim3(n_blocks),
+                         dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
+                         mse_returns, loss_partials);
+    return;
+  }
   if (rows == 16) {
     if (compute_bf16)
       hipLaunchKernelGGL((mlp_bwd_fused_f32_t<16, true>), dim3(n_blocks),
