@@ -370,6 +370,53 @@ class TestFusedValueBackward:
                            [g for pair in zip(grads[1:1+n], grads[1+n:1+2*n]) for g in pair]):
             torch.testing.assert_close(dw, p_e.grad, rtol=5e-4, atol=5e-5)
 
+    @pytest.mark.parametrize("batch", [100, 4000, 9000])
+    def test_fwd_in_kernel_adam_step_matches_separate(self, ext, batch):
+        """DO_FWD + merged reduce/Adam (2-kernel value iteration) must
+        produce the same post-step params and loss partials as the
+        separate fwd -> bwd -> reduce -> FusedAdam chain."""
+        from rl_replicas_amd import ops as _ops
+        from rl_replicas_amd.networks import MLP
+        from rl_replicas_amd.ops.fused_adam import FusedAdam, adam_arg_lists
+        from rl_replicas_amd.ops.fused_mlp import _extract_layers
+
+        torch.manual_seed(5)
+        mlp_f = MLP([17, 64, 32, 1]).to("cuda")
+        mlp_s = MLP([17, 64, 32, 1]).to("cuda")
+        mlp_s.load_state_dict(mlp_f.state_dict())
+        obs = torch.randn(batch, 17, device="cuda")
+        returns = torch.randn(batch, device="cuda")
+
+        # separate-path oracle: fused fwd + value bwd + FusedAdam step
+        w_s, b_s, acts = _extract_layers(mlp_s)
+        opt_s = FusedAdam(mlp_s.parameters(), lr=1e-3)
+        outs = ext.mlp_forward(obs, list(w_s), list(b_s), acts, True)
+        grads = ext.value_mlp_backward(obs, list(w_s), list(b_s),
+                                       list(outs[1:]), outs[0], acts, returns)
+        n = len(w_s)
+        for w, dw in zip(w_s, grads[1 : 1 + n]):
+            w.grad = dw
+        for b, db in zip(b_s, grads[1 + n : 1 + 2 * n]):
+            b.grad = db
+        opt_s.step()
+
+        # merged path: ONE fwd+bwd kernel + ONE reduce+adam kernel
+        w_f, b_f, _ = _extract_layers(mlp_f)
+        opt_f = FusedAdam(mlp_f.parameters(), lr=1e-3)
+        m, v, step0, hp = adam_arg_lists(opt_f, w_f, b_f)
+        dummy = torch.empty(0, device="cuda")
+        out = ext.value_mlp_backward(obs, list(w_f), list(b_f), [], dummy,
+                                     acts, returns, 0, None, m, v, step0,
+                                     *hp, 0.0, True)
+        opt_f.bump_steps(1.0)
+
+        torch.testing.assert_close(out[-1][0], grads[-1][0], rtol=1e-5, atol=1e-7)
+        for p_f, p_s in zip(mlp_f.parameters(), mlp_s.parameters()):
+            torch.testing.assert_close(p_f, p_s, rtol=1e-5, atol=1e-7)
+        s_f = opt_f.state[next(iter(mlp_f.parameters()))]["step"]
+        s_s = opt_s.state[next(iter(mlp_s.parameters()))]["step"]
+        assert float(s_f) == float(s_s) == 1.0
+
 
 class TestSplitGraphs:
     def test_split_graphs_equal_combined(self, ext, monkeypatch):
