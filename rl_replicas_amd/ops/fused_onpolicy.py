@@ -494,7 +494,16 @@ class _GraphedValueLoop:
                 vf.optimizer, weights0, biases0
             )
 
-            fp32 = ops.compute_bf16() == 0
+            # DO_FWD stages L extra activation tiles in LDS — mirror
+            # the C++ budget gate so deep narrow nets fall back to the
+            # separate-forward form instead of tripping the TORCH_CHECK
+            brows = 16 if obs0.shape[0] < 8192 else 32
+            whole_w = sum(
+                w.shape[0] * (w.shape[1] + 1) for w in _extract_layers(mlp)[0]
+            )
+            lds_ok = ((3 + len(_extract_layers(mlp)[0])) * brows * 68
+                      + whole_w) * 4 <= 100 * 1024
+            fp32 = ops.compute_bf16() == 0 and lds_ok
             dummy = torch.empty(0, device=obs0.device)
 
             def body():
