@@ -81,6 +81,22 @@ class FusedAdam(torch.optim.Adam):
         return loss
 
     @torch.no_grad()
+    def bump_steps_by(self, amount: Tensor) -> None:
+        """Advance every step counter by a DEVICE scalar (the captured
+        gated policy loop's executed-iteration count)."""
+        from rl_replicas_amd import ops
+
+        ext = ops._load_extension()
+        steps = [
+            self.state[p]["step"]
+            for group in self.param_groups
+            for p in group["params"]
+            if p in self.state and "step" in self.state[p]
+        ]
+        if steps:
+            ext.adam_bump_dev_(steps, amount)
+
+    @torch.no_grad()
     def bump_steps(self, amount: float) -> None:
         """Advance every step counter by `amount` (one kernel)."""
         from rl_replicas_amd import ops

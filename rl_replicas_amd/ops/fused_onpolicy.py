@@ -344,7 +344,12 @@ class _GraphedPPO:
                                                 acts)
                         if i == 0:
                             self.loss0.copy_(scalars[:1])
-                        policy.optimizer.step(gate=self.gate)
+                        # deferred bias correction: iteration i executed
+                        # iff no stop before it, so step_delta=i is its
+                        # exact prior-update count; the epilogue bumps by
+                        # the device-side executed count once
+                        policy.optimizer.step(step_delta=float(i),
+                                              do_bump=False, gate=self.gate)
 
                 return chunk
 
@@ -353,6 +358,8 @@ class _GraphedPPO:
                 kl = kl_eval()
                 ext.ppo_gate_update_(self.gate, kl, self.kl_final,
                                      self.iters_done, thr)
+                # ... after which iters_done IS the executed-update count
+                policy.optimizer.bump_steps_by(self.iters_done)
 
             # chunked capture: the KL early stop is common in steady
             # state, and a gate-frozen iteration still executes its

@@ -27,9 +27,11 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
                           hipStream_t stream, int rows, int do_fwd = 0);
-// 16-row tiles below 8K rows (measured faster: more WGs hide latency),
-// 32-row beyond — MUST stay consistent with value_loss_partials_blocks
-inline int bwd_fused_rows(int batch) { return batch < 8192 ? 16 : 32; }
+// Backward tiles stay at 32 rows: 16-row tiles made stage-1 faster but
+// DOUBLED the partial-row count, pushing the (latency-bound) reduce
+// from 10.3 to 17.8 us — a measured net loss.  (The FORWARD still uses
+// 16-row tiles below 8K rows; its cost has no reduce to pay.)
+inline int bwd_fused_rows(int batch) { (void)batch; return 32; }
 __global__ void mlp_grad_reduce_onepass_f32(ReduceAllArgs a);
 __global__ void mlp_grad_reduce_adam_f32(ReduceAdamArgs a);
 void launch_mlp_layer_fwd_wide(const float* x, const float* W, const float* B,
@@ -853,6 +855,22 @@ void fused_adam_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> g
   }
 }
 
+__global__ void adam_step_bump_dev_kernel(AdamArgs a, const float* amount);
+
+// bump by a device scalar (captured gated policy loop)
+void adam_bump_dev_(std::vector<torch::Tensor> steps, torch::Tensor amount) {
+  size_t n = steps.size();
+  for (size_t base = 0; base < n; base += MT_MAX_TENSORS) {
+    AdamArgs a{};
+    a.n_tensors = (int)std::min((size_t)MT_MAX_TENSORS, n - base);
+    for (int i = 0; i < a.n_tensors; ++i)
+      a.step[i] = steps[base + i].data_ptr<float>();
+    hipLaunchKernelGGL(adam_step_bump_dev_kernel, dim3(1), dim3(MT_MAX_TENSORS),
+                       0, current_stream(), a, amount.data_ptr<float>());
+    HIP_OK(hipGetLastError());
+  }
+}
+
 // bump-only entry (captured value loop: ONE bump of +num_iters per replay)
 void adam_bump_(std::vector<torch::Tensor> steps, double amount) {
   size_t n = steps.size();
@@ -1121,6 +1139,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("value_loss_finalize", &value_loss_finalize,
         "batched deferred value-loss finalize (gfx950)");
   m.def("adam_bump_", &adam_bump_, "bump Adam step counters (gfx950)");
+  m.def("adam_bump_dev_", &adam_bump_dev_,
+        "bump Adam step counters by a device scalar (gfx950)");
   m.def("ppo_gate_update_", &ppo_gate_update_,
         "device-side KL early-stop gate bookkeeping (gfx950)");
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");

@@ -55,6 +55,14 @@ __global__ void adam_step_bump_kernel(AdamArgs a) {
   if (t < a.n_tensors) a.step[t][0] += a.step_delta;  // 1.0 for a plain step
 }
 
+// bump by a DEVICE scalar (captured gated loops: the executed-iteration
+// count is only known on device)
+__global__ void adam_step_bump_dev_kernel(AdamArgs a,
+                                          const float* __restrict__ amount) {
+  const int t = threadIdx.x;
+  if (t < a.n_tensors) a.step[t][0] += *amount;
+}
+
 // One-thread bookkeeping for the captured PPO policy loop's device-side
 // KL early stop: while the gate is open, record the KL and iteration
 // count; close the gate when the KL crosses the threshold.  Replaces a
