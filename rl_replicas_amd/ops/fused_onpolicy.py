@@ -317,11 +317,55 @@ class _GraphedPPO:
             self.loss0 = torch.zeros(1, device=dev)
             self.iters_done = torch.zeros(1, device=dev)
 
+            # mega path: ONE 3-kernel launch sequence per iteration
+            # (DO_FWD fused fwd+KL+loss+bwd, gate reduce, merged
+            # reduce+Adam incl. the log_std slot) — narrow fp32
+            # identity-head Gaussian policies (the bench config)
+            use_mega = False
+            if kind == "gaussian" and ops.compute_bf16() == 0:
+                weights0, biases0, acts0 = _extract_layers(mlp)
+                brows0 = 32
+                whole_w0 = sum(w.shape[0] * (w.shape[1] + 1) for w in weights0)
+                use_mega = (
+                    acts0[-1] == 0  # identity head
+                    and len(weights0) + 1 <= 5
+                    and policy.log_std.numel() <= 64
+                    and max([obs0.shape[1]] + [w.shape[0] for w in weights0]) <= 64
+                    and ((3 + len(weights0)) * brows0 * 68 + whole_w0) * 4
+                    <= 100 * 1024
+                )
+            if use_mega:
+                from rl_replicas_amd.ops.fused_adam import adam_arg_lists
+
+                weights0, biases0, acts0 = _extract_layers(mlp)
+                m2l, v2l, step0, hp = adam_arg_lists(
+                    policy.optimizer, weights0, biases0
+                )
+                ls_state = policy.optimizer.state[policy.log_std]
+                if len(ls_state) == 0:
+                    ls_state["step"] = torch.zeros((), dtype=torch.float32, device=dev)
+                    ls_state["exp_avg"] = torch.zeros_like(policy.log_std)
+                    ls_state["exp_avg_sq"] = torch.zeros_like(policy.log_std)
+                ls_m, ls_v = ls_state["exp_avg"], ls_state["exp_avg_sq"]
+                scratch_loss = torch.zeros(1, device=dev)
+
             def make_chunk(start: int, count: int):
                 def chunk():
                     if start == 0:
                         self.gate.fill_(1.0)
                         self.iters_done.zero_()
+                    if use_mega:
+                        for i in range(start, start + count):
+                            ext.gaussian_ppo_policy_iter(
+                                self.obs, list(weights0), list(biases0), acts0,
+                                self.actions, self.old_logp, self.adv,
+                                policy.log_std.data, clip, m2l, v2l, ls_m,
+                                ls_v, step0, *hp, float(i), self.gate,
+                                self.kl_final, self.iters_done, thr,
+                                self.loss0 if i == 0 else scratch_loss,
+                                i == 0,
+                            )
+                        return
                     for i in range(start, start + count):
                         # iteration i's saved forward evaluates the policy
                         # at params_i — which is EXACTLY the network output

@@ -26,7 +26,8 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
-                          hipStream_t stream, int rows, int do_fwd = 0);
+                          hipStream_t stream, int rows, int do_fwd = 0,
+                          GaussSeedArgs gargs = GaussSeedArgs{});
 // Backward tiles stay at 32 rows: 16-row tiles made stage-1 faster but
 // DOUBLED the partial-row count, pushing the (latency-bound) reduce
 // from 10.3 to 17.8 us — a measured net loss.  (The FORWARD still uses
@@ -115,6 +116,10 @@ __global__ void fused_polyak_kernel(PolyakArgs a);
 __global__ void ppo_gate_update_kernel(float* gate, const float* kl,
                                        float* kl_final, float* iters_done,
                                        float thr);
+__global__ void ppo_gate_update_reduce_kernel(
+    float* gate, const float* kl_partials, const float* loss_partials,
+    int n_blocks, float inv_b, float* kl_final, float* iters_done, float thr,
+    float* loss_out, int first_iter);
 
 namespace {
 
@@ -1098,6 +1103,137 @@ void ppo_gate_update_(torch::Tensor gate, torch::Tensor kl,
   HIP_OK(hipGetLastError());
 }
 
+// ONE captured Gaussian-PPO policy iteration = 3 kernels:
+//   1. DO_FWD fused kernel: forward (LDS activations) + pending-KL /
+//      loss / dlog_std partials + clipped-surrogate dZ seed + whole-net
+//      backward stage-1
+//   2. gate reduce: close the gate if the pending KL crossed 1.5*maxkl
+//      (so this iteration's update is skipped, like the reference's
+//      break at ppo.py:176-181)
+//   3. merged reduce+Adam over net params AND log_std (pseudo-layer),
+//      gated
+// Narrow fp32 identity-head Gaussian policies only (the bench config).
+torch::Tensor gaussian_ppo_policy_iter(
+    torch::Tensor x, std::vector<torch::Tensor> weights,
+    std::vector<torch::Tensor> biases, std::vector<int64_t> acts,
+    torch::Tensor actions, torch::Tensor old_logp, torch::Tensor advantages,
+    torch::Tensor log_std, double clip, std::vector<torch::Tensor> adam_m,
+    std::vector<torch::Tensor> adam_v, torch::Tensor ls_m, torch::Tensor ls_v,
+    torch::Tensor adam_step, double lr, double beta1, double beta2, double eps,
+    double weight_decay, double step_delta, torch::Tensor gate,
+    torch::Tensor kl_final, torch::Tensor iters_done, double thr,
+    torch::Tensor loss_out, bool first_iter) {
+  const int L = (int)weights.size();
+  check_f32_gpu(x, "x");
+  const int batch = (int)x.size(0);
+  const int D = (int)weights[L - 1].size(0);
+  TORCH_CHECK(acts[L - 1] == 0, "needs an identity head");
+  TORCH_CHECK(L + 1 <= MLP_MAX_LAYERS, "too many layers for the log_std slot");
+  TORCH_CHECK(D <= 64, "action dim must fit one LDS row");
+  int max_width = (int)x.size(1);
+  for (int l = 0; l < L; ++l)
+    max_width = std::max(max_width, (int)weights[l].size(0));
+  TORCH_CHECK(max_width <= 64, "narrow nets only");
+  auto opts = x.options();
+  auto stream = current_stream();
+
+  std::vector<int64_t> totals(L), layer_off(L);
+  int64_t grand = 0;
+  for (int l = 0; l < L; ++l) {
+    totals[l] = (int64_t)weights[l].size(0) * weights[l].size(1) + weights[l].size(0);
+    layer_off[l] = grand;
+    grand += totals[l];
+  }
+  const int64_t grand_ls = grand + D;  // + dlog_std pseudo-layer
+  size_t whole_w = 0;
+  for (auto& w : weights) whole_w += (size_t)w.size(0) * (w.size(1) + 1);
+  const int brows = bwd_fused_rows(batch);
+  const size_t fused_lds = (((size_t)3 + L) * brows * 68 + whole_w) * 4;
+  TORCH_CHECK(fused_lds <= 100 * 1024, "net too large for the fused iter");
+  const int fb = (batch + brows - 1) / brows;
+
+  torch::Tensor ws = torch::empty({(int64_t)fb, grand_ls}, opts);
+  torch::Tensor loss_partials = torch::empty({fb}, opts);
+  torch::Tensor kl_partials = torch::empty({fb}, opts);
+  torch::Tensor dx = torch::empty({batch, x.size(1)}, opts);
+
+  MLPBwdArgs ba{};
+  ba.n_layers = L;
+  ba.batch = batch;
+  ba.ws_stride = grand_ls;
+  ba.dims[0] = (int)x.size(1);
+  for (int l = 0; l < L; ++l) {
+    ba.w[l] = weights[l].data_ptr<float>();
+    ba.h[l] = nullptr;
+    ba.b[l] = biases[l].data_ptr<float>();
+    ba.dims[l + 1] = (int)weights[l].size(0);
+    ba.acts[l] = (int)acts[l];
+    ba.layer_off[l] = (int)layer_off[l];
+  }
+  GaussSeedArgs ga{};
+  ga.actions = actions.data_ptr<float>();
+  ga.old_logp = old_logp.data_ptr<float>();
+  ga.adv = advantages.data_ptr<float>();
+  ga.log_std = log_std.data_ptr<float>();
+  ga.kl_partials = kl_partials.data_ptr<float>();
+  ga.dls_off = (int)grand;
+  ga.clip = (float)clip;
+
+  launch_mlp_bwd_fused(ba, x.data_ptr<float>(), nullptr, dx.data_ptr<float>(),
+                       ws.data_ptr<float>(), nullptr,
+                       loss_partials.data_ptr<float>(), fused_lds, fb, 0,
+                       stream, brows, 1, ga);
+  HIP_OK(hipGetLastError());
+
+  hipLaunchKernelGGL(ppo_gate_update_reduce_kernel, dim3(1), dim3(1), 0,
+                     stream, gate.data_ptr<float>(),
+                     kl_partials.data_ptr<float>(),
+                     loss_partials.data_ptr<float>(), fb, 1.f / (float)batch,
+                     kl_final.data_ptr<float>(), iters_done.data_ptr<float>(),
+                     (float)thr, loss_out.data_ptr<float>(),
+                     first_iter ? 1 : 0);
+  HIP_OK(hipGetLastError());
+
+  TORCH_CHECK((int)adam_m.size() == 2 * L && (int)adam_v.size() == 2 * L);
+  ReduceAdamArgs ra{};
+  ra.ws = ws.data_ptr<float>();
+  ra.stride = grand_ls;
+  ra.n_layers = L + 1;
+  ra.n_blocks = fb;
+  for (int l = 0; l < L; ++l) {
+    ra.pw[l] = weights[l].data_ptr<float>();
+    ra.pb[l] = biases[l].data_ptr<float>();
+    ra.mw[l] = adam_m[l].data_ptr<float>();
+    ra.mb[l] = adam_m[L + l].data_ptr<float>();
+    ra.vw[l] = adam_v[l].data_ptr<float>();
+    ra.vb[l] = adam_v[L + l].data_ptr<float>();
+    ra.total[l] = (int)totals[l];
+    ra.wsize[l] = (int)(weights[l].size(0) * weights[l].size(1));
+  }
+  // log_std pseudo-layer: bias-only (wsize 0)
+  ra.pw[L] = nullptr;
+  ra.pb[L] = log_std.data_ptr<float>();
+  ra.mw[L] = nullptr;
+  ra.mb[L] = ls_m.data_ptr<float>();
+  ra.vw[L] = nullptr;
+  ra.vb[L] = ls_v.data_ptr<float>();
+  ra.total[L] = D;
+  ra.wsize[L] = 0;
+  ra.step = adam_step.data_ptr<float>();
+  ra.lr = (float)lr;
+  ra.beta1 = (float)beta1;
+  ra.beta2 = (float)beta2;
+  ra.eps = (float)eps;
+  ra.weight_decay = (float)weight_decay;
+  ra.step_delta = (float)step_delta;
+  ra.gate = gate.data_ptr<float>();
+  int rb = (int)std::min<int64_t>(256, (grand_ls + 63) / 64);
+  hipLaunchKernelGGL(mlp_grad_reduce_adam_f32, dim3(rb), dim3(256), 0, stream,
+                     ra);
+  HIP_OK(hipGetLastError());
+  return dx;
+}
+
 void counter_add_(torch::Tensor ctr, int64_t delta) {
   TORCH_CHECK(ctr.scalar_type() == torch::kInt64 && ctr.is_cuda() && ctr.numel() == 1,
               "ctr must be a 1-element int64 CUDA tensor");
@@ -1143,6 +1279,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "bump Adam step counters by a device scalar (gfx950)");
   m.def("ppo_gate_update_", &ppo_gate_update_,
         "device-side KL early-stop gate bookkeeping (gfx950)");
+  m.def("gaussian_ppo_policy_iter", &gaussian_ppo_policy_iter,
+        "one 3-kernel Gaussian-PPO policy iteration (gfx950)");
   m.def("segmented_gae", &segmented_gae, "segmented GAE+returns scan (gfx950)");
   m.def("normalize", &normalize, "fused mean/std normalize (gfx950)");
   m.def("q_target", &q_target, "fused Q-learning target (gfx950)");

@@ -38,6 +38,27 @@ DEV_INLINE float act_grad_from_y(int code, float y) {
   }
 }
 
+// PPO/VPG gradient coefficient wrt logp for one row (caller scales by
+// 1/B):  mode 0 = VPG (-A), mode 1 = PPO clipped surrogate.  Gradient
+// semantics replicate torch exactly (min-tie 0.5/0.5, clamp-inclusive).
+#define LOG_2PI 1.8378770664093453f
+DEV_INLINE float dlogp_coeff(int mode, float logp, float old_logp, float adv,
+                             float clip, float* loss_out) {
+  if (mode == 0) {
+    *loss_out = -logp * adv;
+    return -adv;
+  }
+  const float ratio = __expf(logp - old_logp);
+  const float lo = 1.f - clip, hi = 1.f + clip;
+  const float rc = fminf(fmaxf(ratio, lo), hi);
+  const float s1 = ratio * adv;
+  const float s2 = rc * adv;
+  *loss_out = -fminf(s1, s2);
+  const float g1 = (s1 < s2) ? 1.f : (s1 == s2 ? 0.5f : 0.f);
+  const float inclip = (ratio >= lo && ratio <= hi) ? 1.f : 0.f;
+  return -(g1 * s1 + (1.f - g1) * inclip * s1);
+}
+
 // wave-level f32 sum (64 lanes)
 DEV_INLINE float wave_reduce_sum(float v) {
   #pragma unroll
@@ -137,6 +158,22 @@ struct ReduceAdamArgs {
   int n_layers, n_blocks;
   float lr, beta1, beta2, eps, weight_decay, step_delta;
   const float* gate;  // optional: skip entirely while *gate == 0
+};
+
+// Gaussian-PPO seed for the DO_FWD fused backward (mlp_kernels.hip):
+// when `actions` is non-null the dZ seed of the last (identity-head)
+// layer is the PPO clipped-surrogate gradient computed from the
+// LDS-resident forward output, and the kernel additionally emits
+// per-block loss partials, pending-KL partials, and dlog_std partials
+// (written into the stage-1 workspace at dls_off as a pseudo-layer).
+struct GaussSeedArgs {
+  const float* actions;   // [B, D]; nullptr -> plain dy/mse seed
+  const float* old_logp;  // [B]
+  const float* adv;       // [B]
+  const float* log_std;   // [D]
+  float* kl_partials;     // [n_blocks]
+  int dls_off;            // flat ws offset of the dlog_std pseudo-layer
+  float clip;
 };
 
 // replay-ring minibatch gather (offpolicy_kernels.hip / bindings.hip)

@@ -19,7 +19,6 @@
 #include "common.h"
 
 #define LOSS_THREADS 1024
-#define LOG_2PI 1.8378770664093453f
 
 // deterministic block-sum over LOSS_THREADS threads; returns total on
 // every thread (LDS wave partials summed in fixed order)
@@ -33,28 +32,6 @@ DEV_INLINE float block_sum(float v, float* red /*[LOSS_THREADS/WAVE]*/) {
   for (int w = 0; w < LOSS_THREADS / WAVE; ++w) total += red[w];
   __syncthreads();
   return total;
-}
-
-// PPO/VPG gradient coefficient wrt logp for one row (already /B-scaled
-// by the caller):  mode 0 = VPG (-A), mode 1 = PPO clipped surrogate.
-DEV_INLINE float dlogp_coeff(int mode, float logp, float old_logp, float adv,
-                             float clip, float* loss_out) {
-  if (mode == 0) {
-    *loss_out = -logp * adv;
-    return -adv;
-  }
-  const float ratio = __expf(logp - old_logp);
-  const float lo = 1.f - clip, hi = 1.f + clip;
-  const float rc = fminf(fmaxf(ratio, lo), hi);
-  const float s1 = ratio * adv;
-  const float s2 = rc * adv;
-  *loss_out = -fminf(s1, s2);
-  // torch.min tie rule: grad 0.5/0.5 at s1==s2; clamp passes gradient
-  // at the boundaries inclusive
-  const float g1 = (s1 < s2) ? 1.f : (s1 == s2 ? 0.5f : 0.f);
-  const float inclip = (ratio >= lo && ratio <= hi) ? 1.f : 0.f;
-  // ds1/dlogp = ratio*adv = s1; ds2/dlogp = inclip * ratio*adv
-  return -(g1 * s1 + (1.f - g1) * inclip * s1);
 }
 
 // ---------------------------------------------------------------------------

@@ -340,7 +340,8 @@ template <int ROWS, bool BF16 = false, bool DO_FWD = false>
 __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
     MLPBwdArgs args, const float* __restrict__ x, const float* __restrict__ dy,
     float* __restrict__ dx_out, float* __restrict__ workspace,
-    const float* __restrict__ mse_returns, float* __restrict__ loss_partials) {
+    const float* __restrict__ mse_returns, float* __restrict__ loss_partials,
+    GaussSeedArgs gargs = GaussSeedArgs{}) {
   constexpr int MAXW = 64;
   constexpr int LDSW = MAXW + 4;
   constexpr int RT = ROWS / 16;
@@ -408,6 +409,79 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
       fin = hl;
     }
   }
+  // Gaussian-PPO seed (DO_FWD only): loss + pending-KL + dlog_std
+  // partials and the dmean dZ seed, all from the LDS-resident forward
+  if (DO_FWD && gargs.actions != nullptr) {
+    const float* outl = hlds + (L - 1) * ROWS * LDSW;
+    const int D = args.dims[L];
+    __shared__ float cbuf[64];       // per-row dlogp coeff (ROWS <= 64)
+    __shared__ float lbuf[64];       // per-row loss
+    __shared__ float kbuf[64];       // per-row pending KL
+    const float inv_b = 1.f / (float)args.batch;
+    if (tid < ROWS) {
+      const int row = row0 + tid;
+      float lossr = 0.f, klr = 0.f, c = 0.f;
+      if (row < args.batch) {
+        float base = 0.5f * (float)D * LOG_2PI;
+        float q = 0.f;
+        for (int d = 0; d < D; ++d) {
+          const float ls = gargs.log_std[d];
+          base += ls;
+          const float sg = __expf(ls);
+          const float diff =
+              gargs.actions[(long)row * D + d] - outl[tid * LDSW + d];
+          const float z = diff / sg;
+          q += z * z;
+        }
+        const float logp = -0.5f * q - base;
+        klr = gargs.old_logp[row] - logp;
+        float lr_;
+        c = dlogp_coeff(1, logp, gargs.old_logp[row], gargs.adv[row],
+                        gargs.clip, &lr_) * inv_b;
+        lossr = lr_ * inv_b;
+      }
+      cbuf[tid] = c;
+      lbuf[tid] = lossr;
+      kbuf[tid] = klr;
+    }
+    __syncthreads();
+    // dZ seed = dmean (identity head)
+    for (int idx = tid; idx < ROWS * D; idx += 256) {
+      const int r = idx / D, d = idx % D;
+      const int row = row0 + r;
+      float v = 0.f;
+      if (row < args.batch) {
+        const float inv_s2 = __expf(-2.f * gargs.log_std[d]);
+        const float diff =
+            gargs.actions[(long)row * D + d] - outl[r * LDSW + d];
+        v = cbuf[r] * diff * inv_s2;
+      }
+      dza[r * LDSW + d] = v;
+    }
+    // dlog_std partials -> ws pseudo-layer (fixed row order: deterministic)
+    if (tid < D) {
+      const float inv_s2 = __expf(-2.f * gargs.log_std[tid]);
+      float acc = 0.f;
+      for (int r = 0; r < ROWS; ++r) {
+        const int row = row0 + r;
+        if (row < args.batch) {
+          const float diff =
+              gargs.actions[(long)row * D + tid] - outl[r * LDSW + tid];
+          acc += cbuf[r] * (diff * diff * inv_s2 - 1.f);
+        }
+      }
+      wsp[gargs.dls_off + tid] = acc;
+    }
+    if (tid == 0) {
+      float sl = 0.f, sk = 0.f;
+      for (int r = 0; r < ROWS; ++r) {
+        sl += lbuf[r];
+        sk += kbuf[r];
+      }
+      loss_partials[blockIdx.x] = sl;
+      gargs.kl_partials[blockIdx.x] = sk;
+    }
+  } else
   // seed dZ for the last layer: dY * act'(final out), or the fused
   // value-MSE gradient when mse_returns is given
   {
@@ -805,16 +879,17 @@ void launch_mlp_bwd_fused(const MLPBwdArgs& args, const float* x,
                           const float* dy, float* dx, float* ws,
                           const float* mse_returns, float* loss_partials,
                           size_t lds_bytes, int n_blocks, int compute_bf16,
-                          hipStream_t stream, int rows, int do_fwd) {
+                          hipStream_t stream, int rows, int do_fwd,
+                          GaussSeedArgs gargs) {
   if (do_fwd) {  // fp32 only (host gates)
     if (rows == 16)
       hipLaunchKernelGGL((mlp_bwd_fused_f32_t<16, false, true>), dim3(n_blocks),
                          dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
-                         mse_returns, loss_partials);
+                         mse_returns, loss_partials, gargs);
     else
       hipLaunchKernelGGL((mlp_bwd_fused_f32_t<32, false, true>), dim3(n_blocks),
                          dim3(256), lds_bytes, stream, args, x, dy, dx, ws,
-                         mse_returns, loss_partials);
+                         mse_returns, loss_partials, gargs);
     return;
   }
   if (rows == 16) {

@@ -63,6 +63,32 @@ __global__ void adam_step_bump_dev_kernel(AdamArgs a,
   if (t < a.n_tensors) a.step[t][0] += *amount;
 }
 
+// Gate bookkeeping fed by the mega policy-iteration kernel's per-block
+// partials: sums loss and pending-KL partials (fixed order), optionally
+// records the iteration-0 loss, then applies the gate rule.  first_iter
+// has no pending KL (old == current params), so it only stores the loss.
+__global__ void ppo_gate_update_reduce_kernel(
+    float* gate, const float* kl_partials, const float* loss_partials,
+    int n_blocks, float inv_b, float* kl_final, float* iters_done, float thr,
+    float* loss_out, int first_iter) {
+  if (threadIdx.x != 0) return;
+  float sk = 0.f, sl = 0.f;
+  for (int p = 0; p < n_blocks; ++p) {
+    sk += kl_partials[p];
+    sl += loss_partials[p];
+  }
+  if (first_iter) {
+    *loss_out = sl;
+    return;  // no pending KL at iteration 0
+  }
+  const float kl = sk * inv_b;
+  if (*gate != 0.f) {
+    *kl_final = kl;
+    *iters_done += 1.f;
+    if (kl > thr) *gate = 0.f;
+  }
+}
+
 // One-thread bookkeeping for the captured PPO policy loop's device-side
 // KL early stop: while the gate is open, record the KL and iteration
 // count; close the gate when the KL crosses the threshold.  Replaces a
