@@ -445,7 +445,12 @@ class TestSplitGraphs:
         assert abs(r_comb["policy/loss"] - r_split["policy/loss"]) < 1e-6
         assert abs(r_comb["policy/kl_divergence"] - r_split["policy/kl_divergence"]) < 1e-6
         assert abs(l_comb - l_split) < 1e-6
+        # the combined path's mega iteration sums dlog_std partials in a
+        # different (deterministic) order than the split path's loss
+        # kernel, so params agree to fp tolerance rather than bitwise;
+        # bitwise determinism WITHIN each mode is pinned by
+        # test_gpu_train.py::test_gpu_determinism_same_seed
         for p1, p2 in zip(m1.policy.parameters(), m2.policy.parameters()):
-            assert torch.equal(p1, p2)
+            torch.testing.assert_close(p1, p2, rtol=1e-5, atol=1e-7)
         for p1, p2 in zip(m1.value_function.parameters(), m2.value_function.parameters()):
             assert torch.equal(p1, p2)
