@@ -411,7 +411,9 @@ class _GraphedPPO:
             # while keeping host syncs at one gate readback per chunk
             # (vs per-iteration replay+sync, or one full-loop graph
             # that always runs all 80)
-            CHUNK = 10
+            import os as _os
+
+            CHUNK = int(_os.environ.get("RL_REPLICAS_AMD_PPO_CHUNK", "10"))
             self.chunks = []
             for start in range(0, num_iters, CHUNK):
                 count = min(CHUNK, num_iters - start)
