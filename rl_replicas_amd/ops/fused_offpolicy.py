@@ -124,14 +124,22 @@ def q_step(q_function, observations: Tensor, actions: Tensor, targets: Tensor,
 
 
 def policy_step(policy, q_function, observations: Tensor, all_reduce_hook,
-                step_delta: Optional[float] = None, adam=None) -> Tensor:
+                step_delta: Optional[float] = None, adam=None,
+                actor_fwd=None) -> Tensor:
     """One deterministic-actor step through a (frozen) critic:
     loss = -mean(Q(s, mu(s)))  (reference ddpg.py:255-273).
-    Returns the loss as a device scalar."""
+    Returns the loss as a device scalar.
+
+    `actor_fwd`: a precomputed _fwd_saved(...) of the actor on
+    `observations` — the captured loop runs it on a side stream
+    overlapped with the critic steps (it depends only on the policy
+    params, which this iteration's critic updates never touch)."""
     ext = ops._load_extension()
     B = observations.shape[0]
     pm = policy.network
-    a_out, a_hidden, a_w, a_b, a_acts = _fwd_saved(pm, observations.contiguous())
+    if actor_fwd is None:
+        actor_fwd = _fwd_saved(pm, observations.contiguous())
+    a_out, a_hidden, a_w, a_b, a_acts = actor_fwd
 
     qin = torch.cat([observations, a_out], dim=-1).contiguous()
     qm = q_function.network
@@ -365,6 +373,16 @@ class _GraphedOffPolicy:
                     main.wait_stream(s1)
                     # critic steps in parallel: q1 on main, q2 on s1
                     s1.wait_stream(main)
+                    actor_fwd = None
+                    if i % delay == 0:
+                        # prefetch the actor forward for this iteration's
+                        # policy step: depends only on policy params +
+                        # obs, both untouched by the critic steps
+                        s2.wait_stream(main)
+                        with torch.cuda.stream(s2):
+                            actor_fwd = _fwd_saved(
+                                algo.policy.network, obs.contiguous()
+                            )
                     self.q_losses[0][i].copy_(
                         q_step(q_fns[0], obs, None, targets, _noop_hook,
                                qin=qin, step_delta=float(i), adam=q_adams[0])
@@ -377,6 +395,8 @@ class _GraphedOffPolicy:
                                        adam=q_adams[1])
                             )
                     main.wait_stream(s1)
+                    if actor_fwd is not None:
+                        main.wait_stream(s2)
                 else:
                     for q, buf_q in zip(q_fns, self.all_q):
                         buf_q[i].copy_(forward_only(q, qin).view(-1))
@@ -389,7 +409,8 @@ class _GraphedOffPolicy:
                 if i % delay == 0:
                     self.pi_losses[pi_k].copy_(
                         policy_step(algo.policy, q_fns[0], obs, _noop_hook,
-                                    step_delta=float(pi_k), adam=pi_adam)
+                                    step_delta=float(pi_k), adam=pi_adam,
+                                    actor_fwd=(actor_fwd if use_streams else None))
                     )
                     ext.fused_polyak_(polyak_src, polyak_dst, rho)
                     pi_k += 1
