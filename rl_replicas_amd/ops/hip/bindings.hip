@@ -454,7 +454,7 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
         // wide layer: 2D-grid dgrad + wgrad kernels
         launch_mlp_bwd_wide(dy.data_ptr<float>(), y.data_ptr<float>(),
                             xin.data_ptr<float>(), weights[l].data_ptr<float>(),
-                            dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
+                            dx.data_ptr<float>(), ws_ptr + (long)layer_off[l] * n_blocks, grand,
                             batch, out_d, in_d, (int)acts[l], stream);
       }
     } else {
@@ -464,7 +464,7 @@ std::vector<torch::Tensor> mlp_backward(torch::Tensor grad_out, torch::Tensor x,
       // merged dgrad + wgrad/bias partials in one kernel
       launch_mlp_bwd_layer(dy.data_ptr<float>(), y.data_ptr<float>(),
                            xin.data_ptr<float>(), weights[l].data_ptr<float>(),
-                           dx.data_ptr<float>(), ws_ptr + layer_off[l], grand,
+                           dx.data_ptr<float>(), ws_ptr + (long)layer_off[l] * n_blocks, grand,
                            batch, out_d, in_d, (int)acts[l], rows, maxw,
                            n_blocks, wmode, lds_bytes, stream);
     }

@@ -220,7 +220,11 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
   const int wr0 = (wave % RT) * 16;
   const int jt0 = (wave / RT) * 16;
   float* wv = wlds + wave * KCHUNK * 18;  // mode-1 slice: [KCHUNK][16+2]
-  float* wsp = workspace + (long)blockIdx.x * ws_stride;
+  // element-major workspace (ws[elem][block]): the latency-bound reduce
+  // reads each element's partials CONTIGUOUSLY; writers scatter instead
+  // (stores don't stall)
+  float* wsp = workspace + blockIdx.x;
+  const long WSN = gridDim.x;
 
   for (int idx = tid; idx < ROWS * out_d; idx += 256) {
     int r = idx / out_d, c = idx % out_d;
@@ -305,7 +309,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
         #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int orow = it * 16 + (lane >> 4) * 4 + r;
-          if (orow < out_d) wsp[(long)orow * in_d + col] = acc[r];
+          if (orow < out_d) wsp[((long)orow * in_d + col) * WSN] = acc[r];
         }
       }
     }
@@ -316,7 +320,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_layer_f32_t(
     float s = 0.f;
     #pragma unroll 4
     for (int r = 0; r < ROWS; ++r) s += dz[r * LDSW + c];
-    wsp[(long)out_d * in_d + c] = s;
+    wsp[((long)out_d * in_d + c) * WSN] = s;
   }
 }
 
@@ -357,7 +361,8 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
   const int row0 = blockIdx.x * ROWS;
   const int wr0 = (wave % RT) * 16;
   const int jt0 = (wave / RT) * 16;
-  float* wsp = workspace + (long)blockIdx.x * args.ws_stride;
+  float* wsp = workspace + blockIdx.x;  // element-major: ws[elem][block]
+  const long WSN = gridDim.x;
   const int L = args.n_layers;
 
   // whole-net padded W image + layer offsets
@@ -470,7 +475,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
           acc += cbuf[r] * (diff * diff * inv_s2 - 1.f);
         }
       }
-      wsp[gargs.dls_off + tid] = acc;
+      wsp[(long)(gargs.dls_off + tid) * WSN] = acc;
     }
     if (tid == 0) {
       float sl = 0.f, sk = 0.f;
@@ -550,7 +555,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
 
     // ---- wgrad partials: dW[i][j] = sum_r dZ[r][i] X[r][j] ----
     const int n_it = (out_d + 15) / 16;
-    float* lw = wsp + args.layer_off[l];
+    float* lw = wsp + (long)args.layer_off[l] * WSN;
     for (int it = wave; it < n_it; it += 4) {
       const int ii = it * 16 + i;
       for (int jt = 0; jt < in_d; jt += 16) {
@@ -578,7 +583,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
           #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const int orow = it * 16 + (lane >> 4) * 4 + r;
-            if (orow < out_d) lw[(long)orow * in_d + col] = acc[r];
+            if (orow < out_d) lw[((long)orow * in_d + col) * WSN] = acc[r];
           }
         }
       }
@@ -588,7 +593,7 @@ __global__ __launch_bounds__(256) void mlp_bwd_fused_f32_t(
       float s = 0.f;
       #pragma unroll 4
       for (int r = 0; r < ROWS; ++r) s += dz_cur[r * LDSW + c];
-      lw[(long)out_d * in_d + c] = s;
+      lw[((long)out_d * in_d + c) * WSN] = s;
     }
 
     // ---- dgrad: dX[b][j] = sum_k dZ[b][k] W[k][j]; for l>0 the next
@@ -795,7 +800,8 @@ DEV_INLINE void wgrad_wide_body(
   const int wave = tid >> 6;
   const int row0 = bx * ROWS;
   const int og0 = by * 64;  // out-col group
-  float* wsp = workspace + (long)blockIdx.x * ws_stride;
+  float* wsp = workspace + bx;  // element-major: ws[elem][block]
+  const long WSN = gridDim.x;
 
   load_tile<LDSW>(xin, xt, row0, batch, in_d, tid, ROWS);
   const int og_w = min(64, out_d - og0);
@@ -827,7 +833,7 @@ DEV_INLINE void wgrad_wide_body(
       #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int orow = og0 + wave * 16 + (lane >> 4) * 4 + r;
-        if (orow < out_d) wsp[(long)orow * in_d + col] = acc[r];
+        if (orow < out_d) wsp[((long)orow * in_d + col) * WSN] = acc[r];
       }
     }
   }
@@ -837,7 +843,7 @@ DEV_INLINE void wgrad_wide_body(
     float s = 0.f;
     #pragma unroll 4
     for (int r = 0; r < ROWS; ++r) s += dz[r * DZW + c];
-    wsp[(long)out_d * in_d + og0 + c] = s;
+    wsp[((long)out_d * in_d + og0 + c) * WSN] = s;
   }
 }
 
@@ -1015,7 +1021,7 @@ __global__ __launch_bounds__(256) void mlp_grad_reduce_adam_f32(ReduceAdamArgs a
     const int g = g0 + e_local;
     float s = 0.f;
     if (g < grand) {
-      for (int p = p0; p < p1; ++p) s += a.ws[p * a.stride + g];
+      for (int p = p0; p < p1; ++p) s += a.ws[(long)g * a.n_blocks + p];
     }
     red[threadIdx.x] = s;
     __syncthreads();
@@ -1060,7 +1066,7 @@ __global__ __launch_bounds__(256) void mlp_grad_reduce_onepass_f32(ReduceAllArgs
     const int g = g0 + e_local;
     float s = 0.f;
     if (g < grand) {
-      for (int p = p0; p < p1; ++p) s += a.ws[p * a.stride + g];
+      for (int p = p0; p < p1; ++p) s += a.ws[(long)g * a.n_blocks + p];
     }
     red[threadIdx.x] = s;
     __syncthreads();
