@@ -92,6 +92,7 @@ def load_reference() -> Dict:
     try:
         import rl_replicas.experience as ref_experience
         import rl_replicas.optimizers.conjugate_gradient_optimizer as ref_cgo
+        import rl_replicas.policies.categorical_policy as ref_cat
         import rl_replicas.policies.deterministic_policy as ref_det
         import rl_replicas.policies.gaussian_policy as ref_gauss
         import rl_replicas.q_function as ref_qf
@@ -109,6 +110,7 @@ def load_reference() -> Dict:
             "TD3": TD3,
             "MLP": MLP,
             "GaussianPolicy": ref_gauss.GaussianPolicy,
+            "CategoricalPolicy": ref_cat.CategoricalPolicy,
             "DeterministicPolicy": ref_det.DeterministicPolicy,
             "ValueFunction": ref_vf.ValueFunction,
             "QFunction": ref_qf.QFunction,
